@@ -1,0 +1,244 @@
+"""Shared building blocks for the diffusion model zoo.
+
+All hot math routes through the ops dispatch layer (ops/__init__.py): fused
+attention, AdaLN-modulated LayerNorm, RMSNorm, RoPE, gated residuals run as
+hand-written gfx950 HIP kernels on GPU and as the fp32-able torch reference
+on CPU. Plain projections stay nn.Linear (hipBLASLt on ROCm).
+
+The architectures mirror the model families the reference node is documented
+to run (README.md:5 of the reference: Z_IMAGE, FLUX.1, WAN2.2, plus the
+SD/SDXL UNets every ComfyUI install has) — the reference contains no model
+code itself; shapes here follow the public architectures.
+"""
+from __future__ import annotations
+
+import math
+from dataclasses import dataclass
+from typing import Optional, Tuple
+
+import torch
+from torch import nn
+
+from .. import ops
+
+
+class MLPEmbedder(nn.Module):
+    """2-layer SiLU MLP used for timestep / vector conditioning."""
+
+    def __init__(self, in_dim: int, hidden_dim: int):
+        super().__init__()
+        self.in_layer = nn.Linear(in_dim, hidden_dim, bias=True)
+        self.out_layer = nn.Linear(hidden_dim, hidden_dim, bias=True)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return self.out_layer(torch.nn.functional.silu(self.in_layer(x)))
+
+
+class RMSNorm(nn.Module):
+    def __init__(self, dim: int):
+        super().__init__()
+        self.scale = nn.Parameter(torch.ones(dim))
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return ops.rms_norm(x, self.scale)
+
+
+class QKNorm(nn.Module):
+    """Per-head-dim RMSNorm on q and k before attention (FLUX-style)."""
+
+    def __init__(self, head_dim: int):
+        super().__init__()
+        self.query_norm = RMSNorm(head_dim)
+        self.key_norm = RMSNorm(head_dim)
+
+    def forward(self, q: torch.Tensor, k: torch.Tensor):
+        return self.query_norm(q), self.key_norm(k)
+
+
+@dataclass
+class ModOut:
+    shift: torch.Tensor
+    scale: torch.Tensor
+    gate: torch.Tensor
+
+
+class Modulation(nn.Module):
+    """vec -> (shift, scale, gate) x {1 or 2} AdaLN parameter sets."""
+
+    def __init__(self, dim: int, double: bool):
+        super().__init__()
+        self.multiplier = 6 if double else 3
+        self.lin = nn.Linear(dim, self.multiplier * dim, bias=True)
+
+    def forward(self, vec: torch.Tensor):
+        out = self.lin(torch.nn.functional.silu(vec))
+        parts = out.chunk(self.multiplier, dim=-1)
+        first = ModOut(*parts[:3])
+        second = ModOut(*parts[3:]) if self.multiplier == 6 else None
+        return first, second
+
+
+def split_heads(x: torch.Tensor, num_heads: int) -> torch.Tensor:
+    """[B, S, H*D] -> [B, H, S, D]"""
+    B, S, _ = x.shape
+    return x.view(B, S, num_heads, -1).transpose(1, 2)
+
+
+def merge_heads(x: torch.Tensor) -> torch.Tensor:
+    """[B, H, S, D] -> [B, S, H*D]"""
+    B, H, S, D = x.shape
+    return x.transpose(1, 2).reshape(B, S, H * D)
+
+
+def rope_2d_table(
+    h: int, w: int, axes_dim: Tuple[int, ...], theta: float = 10000.0,
+    device=None, text_len: int = 0,
+) -> torch.Tensor:
+    """RoPE cos/sin table for text+image joint sequence.
+
+    axes_dim per-axis head-dim shares (FLUX: (16, 56, 56) summing to 128):
+    axis 0 encodes a scalar index (0 for image tokens, position for text),
+    axes 1/2 encode row/column. Text tokens get position ids on axis 0 and
+    zeros elsewhere (FLUX txt_ids convention). Returns [S_txt + h*w, D/2, 2]
+    fp32, precomputed once per replica per resolution on the owning device.
+    """
+    dev = device if device is not None else "cpu"
+    ys, xs = torch.meshgrid(
+        torch.arange(h, device=dev), torch.arange(w, device=dev), indexing="ij"
+    )
+    img_ids = torch.stack(
+        [torch.zeros_like(ys), ys, xs], dim=-1
+    ).reshape(-1, 3)  # [h*w, 3]
+    txt_ids = torch.zeros(text_len, 3, device=dev)
+    if text_len:
+        txt_ids[:, 0] = torch.arange(text_len, device=dev)
+    ids = torch.cat([txt_ids, img_ids.float()], dim=0)  # [S, 3]
+    tables = [
+        ops.rope_freqs(ids[:, ax], axes_dim[ax], theta) for ax in range(len(axes_dim))
+    ]  # each [S, axes_dim[ax]/2, 2]
+    return torch.cat(tables, dim=1)  # [S, D/2, 2]
+
+
+class JointAttention(nn.Module):
+    """Attention core shared by MMDiT blocks: qk-norm + rope + fused attn."""
+
+    def __init__(self, num_heads: int, head_dim: int):
+        super().__init__()
+        self.num_heads = num_heads
+        self.head_dim = head_dim
+        self.scale = 1.0 / math.sqrt(head_dim)
+
+    def forward(self, q, k, v, pe: Optional[torch.Tensor]):
+        if pe is not None:
+            q = ops.rope_apply(q, pe)
+            k = ops.rope_apply(k, pe)
+        out = ops.attention(q, k, v, self.scale)
+        return merge_heads(out)
+
+
+class DoubleStreamBlock(nn.Module):
+    """FLUX-class dual-stream MMDiT block: separate img/txt params, joint
+    attention over the concatenated sequence."""
+
+    def __init__(self, hidden: int, num_heads: int, mlp_ratio: float = 4.0):
+        super().__init__()
+        head_dim = hidden // num_heads
+        mlp_dim = int(hidden * mlp_ratio)
+        self.num_heads = num_heads
+        self.img_mod = Modulation(hidden, double=True)
+        self.img_attn_qkv = nn.Linear(hidden, hidden * 3)
+        self.img_attn_norm = QKNorm(head_dim)
+        self.img_attn_proj = nn.Linear(hidden, hidden)
+        self.img_mlp = nn.Sequential(
+            nn.Linear(hidden, mlp_dim), nn.GELU(approximate="tanh"),
+            nn.Linear(mlp_dim, hidden),
+        )
+        self.txt_mod = Modulation(hidden, double=True)
+        self.txt_attn_qkv = nn.Linear(hidden, hidden * 3)
+        self.txt_attn_norm = QKNorm(head_dim)
+        self.txt_attn_proj = nn.Linear(hidden, hidden)
+        self.txt_mlp = nn.Sequential(
+            nn.Linear(hidden, mlp_dim), nn.GELU(approximate="tanh"),
+            nn.Linear(mlp_dim, hidden),
+        )
+        self.attn = JointAttention(num_heads, head_dim)
+
+    def _qkv(self, x, qkv_layer, norm):
+        q, k, v = qkv_layer(x).chunk(3, dim=-1)
+        q = split_heads(q, self.num_heads)
+        k = split_heads(k, self.num_heads)
+        v = split_heads(v, self.num_heads)
+        q, k = norm(q, k)
+        return q, k, v
+
+    def forward(self, img, txt, vec, pe):
+        img_m1, img_m2 = self.img_mod(vec)
+        txt_m1, txt_m2 = self.txt_mod(vec)
+
+        img_in = ops.layer_norm_mod(img, img_m1.scale, img_m1.shift)
+        txt_in = ops.layer_norm_mod(txt, txt_m1.scale, txt_m1.shift)
+        iq, ik, iv = self._qkv(img_in, self.img_attn_qkv, self.img_attn_norm)
+        tq, tk, tv = self._qkv(txt_in, self.txt_attn_qkv, self.txt_attn_norm)
+
+        # joint sequence: txt first, then img (FLUX convention)
+        q = torch.cat([tq, iq], dim=2)
+        k = torch.cat([tk, ik], dim=2)
+        v = torch.cat([tv, iv], dim=2)
+        attn = self.attn(q, k, v, pe)
+        txt_attn, img_attn = attn[:, : txt.shape[1]], attn[:, txt.shape[1]:]
+
+        img = ops.gate_residual(img, img_m1.gate, self.img_attn_proj(img_attn))
+        img = ops.gate_residual(
+            img, img_m2.gate,
+            self.img_mlp(ops.layer_norm_mod(img, img_m2.scale, img_m2.shift)),
+        )
+        txt = ops.gate_residual(txt, txt_m1.gate, self.txt_attn_proj(txt_attn))
+        txt = ops.gate_residual(
+            txt, txt_m2.gate,
+            self.txt_mlp(ops.layer_norm_mod(txt, txt_m2.scale, txt_m2.shift)),
+        )
+        return img, txt
+
+
+class SingleStreamBlock(nn.Module):
+    """FLUX-class single-stream block: fused qkv+mlp in, attn ∥ mlp, fused out."""
+
+    def __init__(self, hidden: int, num_heads: int, mlp_ratio: float = 4.0):
+        super().__init__()
+        head_dim = hidden // num_heads
+        self.num_heads = num_heads
+        self.mlp_dim = int(hidden * mlp_ratio)
+        self.linear1 = nn.Linear(hidden, hidden * 3 + self.mlp_dim)
+        self.linear2 = nn.Linear(hidden + self.mlp_dim, hidden)
+        self.norm = QKNorm(head_dim)
+        self.modulation = Modulation(hidden, double=False)
+        self.attn = JointAttention(num_heads, head_dim)
+        self.mlp_act = nn.GELU(approximate="tanh")
+
+    def forward(self, x, vec, pe):
+        mod, _ = self.modulation(vec)
+        x_in = ops.layer_norm_mod(x, mod.scale, mod.shift)
+        qkv, mlp_in = self.linear1(x_in).split(
+            [3 * x.shape[-1], self.mlp_dim], dim=-1
+        )
+        q, k, v = qkv.chunk(3, dim=-1)
+        q = split_heads(q, self.num_heads)
+        k = split_heads(k, self.num_heads)
+        v = split_heads(v, self.num_heads)
+        q, k = self.norm(q, k)
+        attn = self.attn(q, k, v, pe)
+        out = self.linear2(torch.cat([attn, self.mlp_act(mlp_in)], dim=-1))
+        return ops.gate_residual(x, mod.gate, out)
+
+
+class LastLayer(nn.Module):
+    """Final AdaLN + projection to patch output."""
+
+    def __init__(self, hidden: int, out_dim: int):
+        super().__init__()
+        self.ada_lin = nn.Linear(hidden, 2 * hidden, bias=True)
+        self.linear = nn.Linear(hidden, out_dim, bias=True)
+
+    def forward(self, x, vec):
+        shift, scale = self.ada_lin(torch.nn.functional.silu(vec)).chunk(2, dim=-1)
+        return self.linear(ops.layer_norm_mod(x, scale, shift))

@@ -1,0 +1,185 @@
+"""Op dispatch: hand-written gfx950 HIP kernels on GPU, torch reference on CPU.
+
+Policy (deliberate, per the framework's native-code contract):
+- Tensors on a HIP device REQUIRE the in-tree native extension `_pa_hip`.
+  If it is not importable, ops raise instead of silently falling back to
+  eager PyTorch — a GPU run must execute our CDNA4 kernels or fail loudly.
+  (Escape hatch for debugging only: PA_ALLOW_EAGER=1.)
+- CPU tensors use ops.reference — the same functions that serve as the fp32
+  ground truth in the kernel numerics tests.
+- Shapes a kernel doesn't cover dispatch to reference WITH a one-time
+  warning, so coverage gaps are visible in logs, not silent.
+
+Plain GEMMs (nn.Linear / conv) go through PyTorch-ROCm's hipBLASLt/MIOpen
+paths — library GEMMs are the sanctioned exception; the fused/bandwidth
+ops here are the hand-written surface.
+"""
+from __future__ import annotations
+
+import glob
+import importlib.util
+import logging
+import math
+import os
+from typing import Optional
+
+import torch
+
+from . import reference
+
+log = logging.getLogger("parallelanything.ops")
+
+_EXT = None
+_EXT_TRIED = False
+_ALLOW_EAGER = os.environ.get("PA_ALLOW_EAGER", "0") == "1"
+_WARNED: set = set()
+
+
+def _load_ext():
+    global _EXT, _EXT_TRIED
+    if _EXT_TRIED:
+        return _EXT
+    _EXT_TRIED = True
+    pkg_dir = os.path.dirname(__file__)
+    cands = glob.glob(os.path.join(pkg_dir, "_pa_hip*.so"))
+    if not cands:
+        return None
+    try:
+        import torch  # noqa: F401  (must be imported before the ext)
+
+        spec = importlib.util.spec_from_file_location("_pa_hip", cands[0])
+        mod = importlib.util.module_from_spec(spec)
+        spec.loader.exec_module(mod)
+        _EXT = mod
+        log.info("loaded HIP extension %s", cands[0])
+    except Exception as err:  # noqa: BLE001
+        log.error("failed to load HIP extension: %r", err)
+        _EXT = None
+    return _EXT
+
+
+def hip_ext():
+    return _load_ext()
+
+
+def hip_available(op: Optional[str] = None) -> bool:
+    ext = _load_ext()
+    if ext is None:
+        return False
+    return op is None or hasattr(ext, op)
+
+
+def _require_ext(op: str):
+    ext = _load_ext()
+    if ext is not None and hasattr(ext, op):
+        return ext
+    if _ALLOW_EAGER:
+        if op not in _WARNED:
+            _WARNED.add(op)
+            log.warning("PA_ALLOW_EAGER: %s falling back to eager on GPU", op)
+        return None
+    raise RuntimeError(
+        f"HIP extension for op '{op}' is not available on this GPU build — "
+        "build it in-tree with `python -m comfyui_parallelanything_amd.ops.build` "
+        "(gfx950). Refusing silent eager fallback."
+    )
+
+
+def _unsupported(op: str, why: str):
+    key = (op, why)
+    if key not in _WARNED:
+        _WARNED.add(key)
+        log.warning("op %s: shape/dtype not covered by HIP kernel (%s); "
+                    "using reference path", op, why)
+
+
+# ---------------------------------------------------------------------------
+# Public ops
+# ---------------------------------------------------------------------------
+
+def rms_norm(x: torch.Tensor, weight=None, eps: float = 1e-6) -> torch.Tensor:
+    if x.is_cuda:
+        ext = _require_ext("rms_norm")
+        if ext is not None:
+            return ext.rms_norm(x.contiguous(),
+                                weight.contiguous() if weight is not None else None,
+                                eps)
+    return reference.rms_norm(x, weight, eps)
+
+
+def layer_norm_mod(x, scale, shift, eps: float = 1e-6) -> torch.Tensor:
+    if x.is_cuda:
+        ext = _require_ext("layer_norm_mod")
+        if ext is not None:
+            if scale.dim() == x.dim() - 1:
+                return ext.layer_norm_mod(
+                    x.contiguous(), scale.contiguous(), shift.contiguous(), eps
+                )
+            _unsupported("layer_norm_mod", "per-token modulation")
+    return reference.layer_norm_mod(x, scale, shift, eps)
+
+
+def gate_residual(residual, gate, x) -> torch.Tensor:
+    if x.is_cuda:
+        ext = _require_ext("gate_residual")
+        if ext is not None and gate.dim() == x.dim() - 1:
+            return ext.gate_residual(residual.contiguous(), gate.contiguous(),
+                                     x.contiguous())
+    return reference.gate_residual(residual, gate, x)
+
+
+def group_norm_silu(x, num_groups: int, weight=None, bias=None,
+                    eps: float = 1e-6) -> torch.Tensor:
+    if x.is_cuda:
+        ext = _require_ext("group_norm_silu")
+        if ext is not None:
+            if x.dim() == 4 and (x.shape[1] % num_groups) == 0:
+                return ext.group_norm_silu(
+                    x.contiguous(), int(num_groups),
+                    weight.contiguous() if weight is not None else None,
+                    bias.contiguous() if bias is not None else None,
+                    eps,
+                )
+            _unsupported("group_norm_silu", f"dim={x.dim()}")
+    return reference.group_norm_silu(x, num_groups, weight, bias, eps)
+
+
+def rope_apply(x, cs) -> torch.Tensor:
+    if x.is_cuda:
+        ext = _require_ext("rope_apply")
+        if ext is not None:
+            return ext.rope_apply(x.contiguous(), cs.contiguous())
+    return reference.rope_apply(x, cs)
+
+
+def rope_freqs(positions, dim: int, theta: float = 10000.0) -> torch.Tensor:
+    # Table precompute: once per replica per resolution, on the owning GPU
+    # (the reference scrubbed these caches off replicas instead —
+    # clear_flux_caches, any_device_parallel.py:166-195).
+    return reference.rope_freqs(positions, dim, theta)
+
+
+def attention(q, k, v, scale: Optional[float] = None) -> torch.Tensor:
+    """Flash-style fused attention, layout [B, H, S, D]."""
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    if q.is_cuda:
+        ext = _require_ext("attn_fwd")
+        if ext is not None:
+            D = q.shape[-1]
+            if q.dtype == torch.bfloat16 and D in (64, 128):
+                return ext.attn_fwd(q.contiguous(), k.contiguous(),
+                                    v.contiguous(), float(scale))
+            _unsupported("attn_fwd", f"dtype={q.dtype}, D={D}")
+            return reference.attention(q, k, v, scale)
+    return reference.attention(q, k, v, scale)
+
+
+def timestep_embedding(t, dim: int, max_period: float = 10000.0,
+                       time_factor: float = 1000.0) -> torch.Tensor:
+    if t.is_cuda:
+        ext = _require_ext("timestep_embedding")
+        if ext is not None:
+            return ext.timestep_embedding(t.contiguous().float(), int(dim),
+                                          float(max_period), float(time_factor))
+    return reference.timestep_embedding(t, dim, max_period, time_factor)

@@ -1,0 +1,127 @@
+"""Engine routing + golden-equality tests on the [cpu,cpu] chain —
+BASELINE config 1 (SD1.5-class, plumbing, no GPU needed)."""
+import pytest
+import torch
+from torch import nn
+
+from comfyui_parallelanything_amd.models.registry import (
+    flux_inputs,
+    make_flux,
+    make_sd15,
+    sd15_inputs,
+)
+from comfyui_parallelanything_amd.parallel.chain import DeviceChain, make_entry
+from comfyui_parallelanything_amd.parallel.engine import (
+    ParallelEngine,
+    WorkerError,
+    install_parallel_forward,
+    uninstall_parallel_forward,
+)
+
+
+def cpu_chain(*pcts):
+    return DeviceChain.from_list([make_entry("cpu", p) for p in pcts])
+
+
+@pytest.fixture(scope="module")
+def sd15():
+    return make_sd15(tiny=True)
+
+
+def test_golden_dp_equals_single_sd15(sd15):
+    """BASELINE config 1: SD1.5-class 50/50 [cpu,cpu] == single device."""
+    x, t, c, kw = sd15_inputs(2, tiny=True)
+    ref = sd15(x, t, context=c, **kw)
+    eng = ParallelEngine(cpu_chain(50, 50), auto_vram_balance=False)
+    eng.setup(sd15)
+    out = eng.forward(x, t, context=c, **kw)
+    assert out.shape == ref.shape
+    torch.testing.assert_close(out, ref, rtol=1e-4, atol=1e-5)
+
+
+def test_golden_chunkwise_bitwise(sd15):
+    """Each parallel chunk is bit-identical to running that chunk alone."""
+    x, t, c, kw = sd15_inputs(4, tiny=True)
+    eng = ParallelEngine(cpu_chain(50, 50), auto_vram_balance=False)
+    eng.setup(sd15)
+    out = eng.forward(x, t, context=c, **kw)
+    chunk0 = sd15(x[:2], t[:2], context=c[:2], **{k: v[:2] for k, v in kw.items()})
+    assert torch.equal(out[:2], chunk0)
+
+
+def test_golden_weighted_split_flux():
+    m = make_flux(tiny=True, dtype=torch.float32)
+    x, t, c, kw = flux_inputs(7, tiny=True, dtype=torch.float32)
+    ref = m(x, t, context=c, **kw)
+    eng = ParallelEngine(cpu_chain(60, 40), auto_vram_balance=False)
+    eng.setup(m)
+    out = eng.forward(x, t, context=c, **kw)
+    torch.testing.assert_close(out, ref, rtol=1e-4, atol=1e-5)
+
+
+def test_routing_lead_only_when_batch_below_devices(sd15):
+    x, t, c, kw = sd15_inputs(2, tiny=True)
+    eng = ParallelEngine(cpu_chain(34, 33, 33), auto_vram_balance=False)
+    eng.setup(sd15)
+    ref = sd15(x, t, context=c, **kw)
+    out = eng.forward(x, t, context=c, **kw)  # batch 2 < 3 devices
+    assert torch.equal(out, ref)
+
+
+def test_routing_split_disabled(sd15):
+    x, t, c, kw = sd15_inputs(4, tiny=True)
+    eng = ParallelEngine(cpu_chain(50, 50), workload_split=False,
+                         auto_vram_balance=False)
+    eng.setup(sd15)
+    out = eng.forward(x, t, context=c, **kw)
+    assert torch.equal(out, sd15(x, t, context=c, **kw))
+
+
+def test_worker_error_attribution():
+    class Boom(nn.Module):
+        def forward(self, x, t, context=None):
+            if x.shape[0] == 1:  # second chunk
+                raise ValueError("boom")
+            return x
+
+    eng = ParallelEngine(cpu_chain(75, 25), auto_vram_balance=False)
+    eng.setup(Boom())
+    with pytest.raises(WorkerError) as ei:
+        eng.forward(torch.zeros(4, 3), torch.zeros(4))
+    assert "cpu" in str(ei.value)
+
+
+def test_setup_oom_drops_device(monkeypatch):
+    from comfyui_parallelanything_amd.parallel import engine as eng_mod
+
+    calls = []
+
+    real = eng_mod.replicate_module
+
+    def fake_replicate(src, dev, force_copy=False, non_blocking=True):
+        calls.append(dev)
+        if len(calls) == 2:  # second device OOMs
+            raise torch.cuda.OutOfMemoryError("fake OOM")
+        return real(src, dev, force_copy, non_blocking)
+
+    monkeypatch.setattr(eng_mod, "replicate_module", fake_replicate)
+    m = nn.Linear(3, 3)
+    eng = eng_mod.ParallelEngine(cpu_chain(40, 40, 20), auto_vram_balance=False)
+    eng.setup(m)
+    assert len(eng.chain.devices) == 2
+    assert eng.chain.weights[0] == pytest.approx(40 / 60)
+
+
+def test_install_uninstall_roundtrip(sd15):
+    eng = ParallelEngine(cpu_chain(50, 50), auto_vram_balance=False)
+    eng.setup(sd15)
+    orig_forward = sd15.forward
+    install_parallel_forward(sd15, eng)
+    assert sd15._true_parallel_active
+    assert sd15._parallel_devices == ("cpu", "cpu")
+    x, t, c, kw = sd15_inputs(2, tiny=True)
+    out = sd15(x, t, context=c, **kw)  # goes through the engine
+    assert out.shape == x.shape
+    uninstall_parallel_forward(sd15)
+    assert not hasattr(sd15, "_true_parallel_active")
+    assert sd15.forward == orig_forward
