@@ -22,8 +22,9 @@ def _latent_hw(px: int) -> int:
 def make_flux(dev="cpu", dtype=torch.bfloat16, tiny=False):
     cfg = FluxConfig.tiny() if tiny else FluxConfig.flux1_dev()
     torch.manual_seed(0)
-    m = Flux(cfg).to(device=dev, dtype=dtype).eval()
-    return m
+    with torch.device(dev):
+        m = Flux(cfg)
+    return m.to(dtype=dtype).eval()
 
 
 def flux_inputs(batch: int, px: int = 1024, dev="cpu", dtype=torch.bfloat16,
@@ -42,7 +43,9 @@ def flux_inputs(batch: int, px: int = 1024, dev="cpu", dtype=torch.bfloat16,
 def make_zimage(dev="cpu", dtype=torch.bfloat16, tiny=False):
     cfg = ZImageConfig.tiny() if tiny else ZImageConfig.z_image_turbo()
     torch.manual_seed(0)
-    return ZImage(cfg).to(device=dev, dtype=dtype).eval()
+    with torch.device(dev):
+        m = ZImage(cfg)
+    return m.to(dtype=dtype).eval()
 
 
 def zimage_inputs(batch: int, px: int = 1024, dev="cpu", dtype=torch.bfloat16,
@@ -60,7 +63,9 @@ def zimage_inputs(batch: int, px: int = 1024, dev="cpu", dtype=torch.bfloat16,
 def make_sd15(dev="cpu", dtype=torch.float32, tiny=False):
     cfg = UNetConfig.tiny() if tiny else UNetConfig.sd15()
     torch.manual_seed(0)
-    return SDUNet(cfg).to(device=dev, dtype=dtype).eval()
+    with torch.device(dev):
+        m = SDUNet(cfg)
+    return m.to(dtype=dtype).eval()
 
 
 def sd15_inputs(batch: int, px: int = 256, dev="cpu", dtype=torch.float32,
@@ -78,7 +83,9 @@ def sd15_inputs(batch: int, px: int = 256, dev="cpu", dtype=torch.float32,
 def make_sdxl(dev="cpu", dtype=torch.bfloat16, tiny=False):
     cfg = UNetConfig.tiny() if tiny else UNetConfig.sdxl()
     torch.manual_seed(0)
-    return SDUNet(cfg).to(device=dev, dtype=dtype).eval()
+    with torch.device(dev):
+        m = SDUNet(cfg)
+    return m.to(dtype=dtype).eval()
 
 
 def sdxl_inputs(batch: int, px: int = 1024, dev="cpu", dtype=torch.bfloat16,
@@ -98,7 +105,9 @@ def sdxl_inputs(batch: int, px: int = 1024, dev="cpu", dtype=torch.bfloat16,
 def make_wan(dev="cpu", dtype=torch.bfloat16, tiny=False):
     cfg = WanConfig.tiny() if tiny else WanConfig.wan22_a14b()
     torch.manual_seed(0)
-    return WanDiT(cfg).to(device=dev, dtype=dtype).eval()
+    with torch.device(dev):
+        m = WanDiT(cfg)
+    return m.to(dtype=dtype).eval()
 
 
 def wan_inputs(batch: int, frames: int = 21, h: int = 90, w: int = 160,

@@ -1,0 +1,665 @@
+// MI355X (gfx950, CDNA4) native kernels for comfyui_parallelanything_amd.
+//
+// Hand-written HIP: fused attention (MFMA 16x16x32 bf16, LDS-tiled),
+// AdaLN-modulated LayerNorm, RMSNorm, GroupNorm+SiLU, RoPE apply, gated
+// residual, timestep embedding. No CUDA-compat shims, no hipify: this file
+// targets gfx950 only (wave64, 4xSIMD-32 CUs, 160 KiB LDS, per-XCD L2).
+//
+// These replace the per-step model math the reference node left to stock
+// PyTorch inside ComfyUI's model implementations: the
+// reference repo has no native code, so every kernel here is new work
+// driven by the behavioral contract of the ops/reference.py functions,
+// which are the fp32 ground truth in tests/test_gpu_kernels.py.
+//
+// Build: ops/build.py (hipcc --offload-arch=gfx950, in-tree .so).
+
+#include <torch/extension.h>
+#include <ATen/ATen.h>
+#include <c10/hip/HIPStream.h>
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <hip/hip_fp16.h>
+
+#define PA_DEV __device__ __forceinline__
+
+using bf16 = __hip_bfloat16;
+typedef __attribute__((ext_vector_type(8))) short short8;   // 8 x bf16 (4 VGPR)
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+typedef __attribute__((ext_vector_type(4))) short short4v;  // 4 x bf16
+typedef __attribute__((ext_vector_type(2))) float f32x2;
+
+static constexpr int WAVE = 64;
+
+PA_DEV float bf2f(bf16 v) { return __bfloat162float(v); }
+PA_DEV bf16 f2bf(float v) { return __float2bfloat16(v); }
+
+// ---------------------------------------------------------------------------
+// Wave/block reductions (wave64; lane groups via __shfl_xor over 64 lanes).
+// ---------------------------------------------------------------------------
+PA_DEV float wave_reduce_sum(float v) {
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) v += __shfl_xor(v, off, 64);
+    return v;
+}
+
+template <int BLOCK>
+PA_DEV float block_reduce_sum(float v, float* scratch) {
+    constexpr int NW = BLOCK / WAVE;
+    const int lane = threadIdx.x & (WAVE - 1);
+    const int wid = threadIdx.x / WAVE;
+    v = wave_reduce_sum(v);
+    __syncthreads();  // protect scratch against the previous reduction's reads
+    if (lane == 0) scratch[wid] = v;
+    __syncthreads();
+    float total = 0.f;
+#pragma unroll
+    for (int i = 0; i < NW; ++i) total += scratch[i];  // LDS broadcast reads
+    return total;
+}
+
+// ---------------------------------------------------------------------------
+// RMSNorm: one 256-thread block per row, vectorized bf16x8 loads (G13).
+// rows = prod(leading dims), D = last dim.
+// ---------------------------------------------------------------------------
+template <typename T>
+__global__ void rms_norm_kernel(const T* __restrict__ x, const T* __restrict__ w,
+                                T* __restrict__ out, int D, float eps) {
+    const long row = blockIdx.x;
+    const T* xr = x + row * (long)D;
+    T* yr = out + row * (long)D;
+    __shared__ float scratch[8];
+
+    float acc = 0.f;
+    for (int i = threadIdx.x; i < D; i += blockDim.x) {
+        float v = (float)xr[i];
+        acc += v * v;
+    }
+    float ssq = block_reduce_sum<256>(acc, scratch);
+    const float rrms = rsqrtf(ssq / (float)D + eps);
+    for (int i = threadIdx.x; i < D; i += blockDim.x) {
+        float v = (float)xr[i] * rrms;
+        if (w != nullptr) v *= (float)w[i];
+        yr[i] = (T)v;
+    }
+}
+
+// bf16 fast path: 8-wide vector loads/stores.
+__global__ void rms_norm_bf16_kernel(const bf16* __restrict__ x,
+                                     const bf16* __restrict__ w,
+                                     bf16* __restrict__ out, int D, float eps) {
+    const long row = blockIdx.x;
+    const short8* xr = reinterpret_cast<const short8*>(x + row * (long)D);
+    short8* yr = reinterpret_cast<short8*>(out + row * (long)D);
+    const short8* wv = reinterpret_cast<const short8*>(w);
+    const int DV = D / 8;
+    __shared__ float scratch[8];
+
+    float acc = 0.f;
+    for (int i = threadIdx.x; i < DV; i += blockDim.x) {
+        short8 v = xr[i];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            float f = bf2f(__ushort_as_bfloat16((unsigned short)v[j]));
+            acc += f * f;
+        }
+    }
+    float ssq = block_reduce_sum<256>(acc, scratch);
+    const float rrms = rsqrtf(ssq / (float)D + eps);
+    for (int i = threadIdx.x; i < DV; i += blockDim.x) {
+        short8 v = xr[i];
+        short8 o;
+        if (w != nullptr) {
+            short8 wv8 = wv[i];
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                float f = bf2f(__ushort_as_bfloat16((unsigned short)v[j])) * rrms *
+                          bf2f(__ushort_as_bfloat16((unsigned short)wv8[j]));
+                o[j] = (short)__bfloat16_as_ushort(f2bf(f));
+            }
+        } else {
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                float f = bf2f(__ushort_as_bfloat16((unsigned short)v[j])) * rrms;
+                o[j] = (short)__bfloat16_as_ushort(f2bf(f));
+            }
+        }
+        yr[i] = o;
+    }
+}
+
+// ---------------------------------------------------------------------------
+// AdaLN-modulated LayerNorm: out = LN(x) * (1 + scale[b]) + shift[b].
+// x: [B, S, D]; scale/shift: [B, D]. One block per (b, s) row.
+// ---------------------------------------------------------------------------
+template <typename T>
+__global__ void layer_norm_mod_kernel(const T* __restrict__ x,
+                                      const T* __restrict__ scale,
+                                      const T* __restrict__ shift,
+                                      T* __restrict__ out,
+                                      int S, int D, float eps) {
+    const long row = blockIdx.x;          // b * S + s
+    const long b = row / S;
+    const T* xr = x + row * (long)D;
+    const T* sc = scale + b * (long)D;
+    const T* sh = shift + b * (long)D;
+    T* yr = out + row * (long)D;
+    __shared__ float scratch[8];
+
+    float s1 = 0.f, s2 = 0.f;
+    for (int i = threadIdx.x; i < D; i += blockDim.x) {
+        float v = (float)xr[i];
+        s1 += v;
+        s2 += v * v;
+    }
+    // two reductions share the scratch sequentially
+    float mean = block_reduce_sum<256>(s1, scratch) / (float)D;
+    float var = block_reduce_sum<256>(s2, scratch) / (float)D - mean * mean;
+    const float rstd = rsqrtf(var + eps);
+    for (int i = threadIdx.x; i < D; i += blockDim.x) {
+        float v = ((float)xr[i] - mean) * rstd;
+        v = v * (1.f + (float)sc[i]) + (float)sh[i];
+        yr[i] = (T)v;
+    }
+}
+
+// ---------------------------------------------------------------------------
+// Gated residual: out = residual + gate[b] * x ;  x,res: [B, S, D], gate [B, D]
+// ---------------------------------------------------------------------------
+template <typename T>
+__global__ void gate_residual_kernel(const T* __restrict__ res,
+                                     const T* __restrict__ gate,
+                                     const T* __restrict__ x,
+                                     T* __restrict__ out,
+                                     long total, int S, int D) {
+    const long stride = (long)gridDim.x * blockDim.x;
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total; i += stride) {
+        const long d = i % D;
+        const long b = i / ((long)S * D);
+        out[i] = (T)((float)res[i] + (float)gate[b * D + d] * (float)x[i]);
+    }
+}
+
+// ---------------------------------------------------------------------------
+// GroupNorm + SiLU: x [B, C, H, W]; one block per (b, group).
+// ---------------------------------------------------------------------------
+template <typename T>
+__global__ void group_norm_silu_kernel(const T* __restrict__ x,
+                                       const float* __restrict__ w,
+                                       const float* __restrict__ bias,
+                                       T* __restrict__ out,
+                                       int C, long HW, int G, float eps) {
+    const int b = blockIdx.x / G;
+    const int g = blockIdx.x % G;
+    const int cpg = C / G;
+    const long base = ((long)b * C + (long)g * cpg) * HW;
+    const long n = (long)cpg * HW;
+    __shared__ float scratch[8];
+
+    float s1 = 0.f, s2 = 0.f;
+    for (long i = threadIdx.x; i < n; i += blockDim.x) {
+        float v = (float)x[base + i];
+        s1 += v;
+        s2 += v * v;
+    }
+    float mean = block_reduce_sum<256>(s1, scratch) / (float)n;
+    float var = block_reduce_sum<256>(s2, scratch) / (float)n - mean * mean;
+    const float rstd = rsqrtf(var + eps);
+    for (long i = threadIdx.x; i < n; i += blockDim.x) {
+        const int c = g * cpg + (int)(i / HW);
+        float v = ((float)x[base + i] - mean) * rstd;
+        if (w != nullptr) v = v * w[c] + bias[c];
+        out[base + i] = (T)(v / (1.f + __expf(-v)));  // SiLU
+    }
+}
+
+// ---------------------------------------------------------------------------
+// RoPE apply: x [B, H, S, D], cs [S, D/2, 2] fp32 -> rotate adjacent pairs.
+// One thread per pair, grid-stride; bf16 pair loaded as one uint.
+// ---------------------------------------------------------------------------
+template <typename T>
+__global__ void rope_apply_kernel(const T* __restrict__ x,
+                                  const float* __restrict__ cs,
+                                  T* __restrict__ out,
+                                  long total_pairs, int S, int Dh) {
+    // Dh = D/2 pairs per row; cs indexed by (s, pair)
+    const long stride = (long)gridDim.x * blockDim.x;
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total_pairs;
+         i += stride) {
+        const long pair = i % Dh;
+        const long s = (i / Dh) % S;
+        const float c = cs[(s * Dh + pair) * 2 + 0];
+        const float sn = cs[(s * Dh + pair) * 2 + 1];
+        const float x0 = (float)x[2 * i];
+        const float x1 = (float)x[2 * i + 1];
+        out[2 * i] = (T)(x0 * c - x1 * sn);
+        out[2 * i + 1] = (T)(x0 * sn + x1 * c);
+    }
+}
+
+// ---------------------------------------------------------------------------
+// Timestep embedding: t [B] -> out [B, dim] fp32 (cos | sin halves).
+// ---------------------------------------------------------------------------
+__global__ void timestep_embedding_kernel(const float* __restrict__ t,
+                                          float* __restrict__ out,
+                                          int B, int dim, float max_period,
+                                          float time_factor) {
+    const int half = dim / 2;
+    const long total = (long)B * half;
+    const long stride = (long)gridDim.x * blockDim.x;
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+         i += stride) {
+        const int b = (int)(i / half);
+        const int j = (int)(i % half);
+        const float freq = __expf(-__logf(max_period) * (float)j / (float)half);
+        const float arg = t[b] * time_factor * freq;
+        out[(long)b * dim + j] = __cosf(arg);
+        out[(long)b * dim + half + j] = __sinf(arg);
+        if ((dim & 1) && j == 0) out[(long)b * dim + dim - 1] = 0.f;
+    }
+}
+
+// ---------------------------------------------------------------------------
+// Fused attention forward (flash-style, non-causal), bf16, D in {64, 128}.
+//
+// Geometry: 256-thread workgroups (4 waves). Each wave owns QBLK=16 query
+// rows; a workgroup covers 64 rows of one (batch, head). K/V stream through
+// LDS in KVBLK=32-key tiles. MFMA v_mfma_f32_16x16x32_bf16 for both QK^T
+// and P*V; online softmax with per-row running (m, l) held across lanes
+// (C-fragment rows live in 16-lane groups; reductions via __shfl_xor<16).
+//
+// LDS layouts (conflict-free by row padding, guide §6 G4):
+//   K tile  : [KVBLK][D + 8]      row stride 272 B (D=128) -> banks disperse
+//   V tile  : [D][KVBLK + 8]      TRANSPOSED at stage time (PV B-fragment
+//                                  wants 8 consecutive keys per lane)
+//   P tile  : per-wave [16][KVBLK + 8]
+// ---------------------------------------------------------------------------
+using bf16x8 = short8;
+
+PA_DEV f32x4 mfma16x16x32(bf16x8 a, bf16x8 b, f32x4 c) {
+    return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+}
+
+template <int D>
+__global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
+    const bf16* __restrict__ q, const bf16* __restrict__ k,
+    const bf16* __restrict__ v, bf16* __restrict__ out,
+    int S, float scale) {
+    constexpr int KVBLK = 32;
+    constexpr int QBLK = 16;          // per wave
+    constexpr int WAVES = 4;
+    constexpr int KPAD = D + 8;       // K row stride (elements)
+    constexpr int VPAD = KVBLK + 8;   // V^T row stride
+    constexpr int PPAD = KVBLK + 8;
+
+    __shared__ bf16 k_lds[KVBLK * KPAD];
+    __shared__ bf16 v_lds[D * VPAD];
+    __shared__ bf16 p_lds[WAVES * QBLK * PPAD];
+
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wid = tid >> 6;
+    const int l16 = lane & 15;        // fragment row / col within 16
+    const int lg = lane >> 4;         // 16-lane group id (0..3)
+
+    // block -> (bh, qtile); q rows for this wave:
+    const long bh = blockIdx.y;
+    const int q0 = blockIdx.x * (WAVES * QBLK) + wid * QBLK;
+    const long base = bh * (long)S * D;
+
+    const bf16* qp = q + base;
+    const bf16* kp = k + base;
+    const bf16* vp = v + base;
+    bf16* op = out + base;
+
+    // ---- load Q fragments (A-layout): lane holds Q[l16][lg*8 + j + 32*kk]
+    constexpr int KK = D / 32;        // MFMA K-steps per QK tile
+    bf16x8 qfrag[KK];
+    {
+        const int row = q0 + l16;
+        const int rr = row < S ? row : S - 1;
+#pragma unroll
+        for (int kk = 0; kk < KK; ++kk) {
+            qfrag[kk] = *reinterpret_cast<const bf16x8*>(
+                qp + (long)rr * D + kk * 32 + lg * 8);
+        }
+        if (row >= S) {
+#pragma unroll
+            for (int kk = 0; kk < KK; ++kk) qfrag[kk] = bf16x8{0,0,0,0,0,0,0,0};
+        }
+    }
+
+    // ---- running state: each lane owns 4 rows (r = lg*4 + rr) x cols l16(+16)
+    constexpr int ND = D / 16;        // PV output col tiles
+    f32x4 o_acc[ND];
+#pragma unroll
+    for (int n = 0; n < ND; ++n) o_acc[n] = f32x4{0.f, 0.f, 0.f, 0.f};
+    float m_run[4], l_run[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) { m_run[r] = -1e30f; l_run[r] = 0.f; }
+
+    bf16* my_p = p_lds + wid * QBLK * PPAD;
+
+    for (int kv0 = 0; kv0 < S; kv0 += KVBLK) {
+        // ---- stage K tile [KVBLK][D] and V^T tile [D][KVBLK] -------------
+        __syncthreads();
+        {
+            // K: 256 threads load KVBLK*D bf16 with 8-elem vectors
+            constexpr int VECS = KVBLK * D / 8;     // 512 (D=128) / 256 (D=64)
+            for (int i = tid; i < VECS; i += 256) {
+                const int row = i / (D / 8);
+                const int col = (i % (D / 8)) * 8;
+                const int src = kv0 + row;
+                bf16x8 val = (src < S)
+                    ? *reinterpret_cast<const bf16x8*>(kp + (long)src * D + col)
+                    : bf16x8{0,0,0,0,0,0,0,0};
+                *reinterpret_cast<bf16x8*>(&k_lds[row * KPAD + col]) = val;
+                // V transposed: element [row][col+j] -> v_lds[(col+j)][row]
+                bf16x8 vv = (src < S)
+                    ? *reinterpret_cast<const bf16x8*>(vp + (long)src * D + col)
+                    : bf16x8{0,0,0,0,0,0,0,0};
+#pragma unroll
+                for (int j = 0; j < 8; ++j)
+                    v_lds[(col + j) * VPAD + row] =
+                        __ushort_as_bfloat16((unsigned short)vv[j]);
+            }
+        }
+        __syncthreads();
+
+        // ---- QK^T: 2 col tiles x KK MFMAs ---------------------------------
+        f32x4 s_acc[2];
+#pragma unroll
+        for (int n = 0; n < 2; ++n) {
+            s_acc[n] = f32x4{0.f, 0.f, 0.f, 0.f};
+            const int key = n * 16 + l16;
+#pragma unroll
+            for (int kk = 0; kk < KK; ++kk) {
+                bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(
+                    &k_lds[key * KPAD + kk * 32 + lg * 8]);
+                s_acc[n] = mfma16x16x32(qfrag[kk], bfrag, s_acc[n]);
+            }
+        }
+
+        // ---- online softmax ----------------------------------------------
+        // s_acc[n][r] = S[q0 + lg*4 + r][kv0 + n*16 + l16] * (pre-scale)
+        // mask tail keys (kv index >= S) to -inf before max/exp
+        const bool key_ok0 = (kv0 + 0 * 16 + l16) < S;
+        const bool key_ok1 = (kv0 + 1 * 16 + l16) < S;
+        float pmax[4];
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            float a = key_ok0 ? s_acc[0][r] * scale : -1e30f;
+            float b = key_ok1 ? s_acc[1][r] * scale : -1e30f;
+            s_acc[0][r] = a; s_acc[1][r] = b;
+            float mx = fmaxf(a, b);
+#pragma unroll
+            for (int off = 1; off < 16; off <<= 1)
+                mx = fmaxf(mx, __shfl_xor(mx, off, 64));
+            pmax[r] = mx;
+        }
+        float alpha[4];
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            const float mnew = fmaxf(m_run[r], pmax[r]);
+            alpha[r] = __expf(m_run[r] - mnew);
+            m_run[r] = mnew;
+            float p0 = __expf(s_acc[0][r] - mnew);
+            float p1 = __expf(s_acc[1][r] - mnew);
+            s_acc[0][r] = p0; s_acc[1][r] = p1;
+            float ps = p0 + p1;
+#pragma unroll
+            for (int off = 1; off < 16; off <<= 1)
+                ps += __shfl_xor(ps, off, 64);
+            l_run[r] = l_run[r] * alpha[r] + ps;
+        }
+        // rescale O accumulator
+#pragma unroll
+        for (int n = 0; n < ND; ++n)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) o_acc[n][r] *= alpha[r];
+
+        // ---- P -> LDS (bf16) then PV --------------------------------------
+        // lane writes P[row = lg*4 + r][key = n*16 + l16]
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            my_p[(lg * 4 + r) * PPAD + l16] = f2bf(s_acc[0][r]);
+            my_p[(lg * 4 + r) * PPAD + 16 + l16] = f2bf(s_acc[1][r]);
+        }
+        __builtin_amdgcn_s_waitcnt(0);   // lgkm drain (wave-local LDS)
+#pragma unroll
+        for (int n = 0; n < ND; ++n) {
+            // A-frag: P[l16][lg*8 + j]; B-frag: V^T[dim = n*16 + l16][key = lg*8 + j]
+            bf16x8 pa = *reinterpret_cast<const bf16x8*>(
+                &my_p[l16 * PPAD + lg * 8]);
+            bf16x8 vb = *reinterpret_cast<const bf16x8*>(
+                &v_lds[(n * 16 + l16) * VPAD + lg * 8]);
+            o_acc[n] = mfma16x16x32(pa, vb, o_acc[n]);
+        }
+    }
+
+    // ---- epilogue: O / l, store -------------------------------------------
+    // lane holds O[row = q0 + lg*4 + r][dim = n*16 + l16]
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+        const int row = q0 + lg * 4 + r;
+        if (row >= S) continue;
+        const float inv_l = (l_run[r] > 0.f) ? 1.f / l_run[r] : 0.f;
+#pragma unroll
+        for (int n = 0; n < ND; ++n) {
+            op[(long)row * D + n * 16 + l16] = f2bf(o_acc[n][r] * inv_l);
+        }
+    }
+}
+
+// ===========================================================================
+// Host-side launchers / torch bindings
+// ===========================================================================
+
+#define CHECK_GPU(t) TORCH_CHECK((t).is_cuda(), #t " must be on a HIP device")
+#define CHECK_LASTDIM(t) TORCH_CHECK((t).stride(-1) == 1, #t " last dim must be contiguous")
+
+static hipStream_t cur_stream() {
+    return c10::hip::getCurrentHIPStream().stream();
+}
+
+at::Tensor rms_norm(at::Tensor x, std::optional<at::Tensor> w, double eps) {
+    CHECK_GPU(x);
+    auto xc = x.contiguous();
+    auto out = at::empty_like(xc);
+    const long rows = xc.numel() / xc.size(-1);
+    const int D = (int)xc.size(-1);
+    const bf16* wp = nullptr;
+    at::Tensor wc;
+    if (w.has_value()) {
+        wc = w->contiguous();
+        TORCH_CHECK(wc.scalar_type() == xc.scalar_type(), "weight dtype mismatch");
+    }
+    if (xc.scalar_type() == at::kBFloat16 && (D % 8) == 0) {
+        hipLaunchKernelGGL(rms_norm_bf16_kernel, dim3((unsigned)rows), dim3(256), 0,
+                           cur_stream(),
+                           (const bf16*)xc.data_ptr(),
+                           w.has_value() ? (const bf16*)wc.data_ptr() : nullptr,
+                           (bf16*)out.data_ptr(), D, (float)eps);
+    } else if (xc.scalar_type() == at::kFloat) {
+        hipLaunchKernelGGL(rms_norm_kernel<float>, dim3((unsigned)rows), dim3(256), 0,
+                           cur_stream(), xc.data_ptr<float>(),
+                           w.has_value() ? wc.data_ptr<float>() : nullptr,
+                           out.data_ptr<float>(), D, (float)eps);
+    } else if (xc.scalar_type() == at::kBFloat16) {
+        hipLaunchKernelGGL(rms_norm_kernel<bf16>, dim3((unsigned)rows), dim3(256), 0,
+                           cur_stream(), (const bf16*)xc.data_ptr(),
+                           w.has_value() ? (const bf16*)wc.data_ptr() : nullptr,
+                           (bf16*)out.data_ptr(), D, (float)eps);
+    } else {
+        TORCH_CHECK(false, "rms_norm: unsupported dtype");
+    }
+    return out;
+}
+
+at::Tensor layer_norm_mod(at::Tensor x, at::Tensor scale, at::Tensor shift,
+                          double eps) {
+    CHECK_GPU(x);
+    TORCH_CHECK(x.dim() == 3, "layer_norm_mod expects [B, S, D]");
+    auto xc = x.contiguous();
+    auto sc = scale.contiguous();
+    auto sh = shift.contiguous();
+    TORCH_CHECK(sc.sizes() == sh.sizes() && sc.dim() == 2, "scale/shift [B, D]");
+    auto out = at::empty_like(xc);
+    const int B = (int)xc.size(0), S = (int)xc.size(1), D = (int)xc.size(2);
+    dim3 grid((unsigned)((long)B * S));
+    if (xc.scalar_type() == at::kBFloat16) {
+        hipLaunchKernelGGL(layer_norm_mod_kernel<bf16>, grid, dim3(256), 0,
+                           cur_stream(), (const bf16*)xc.data_ptr(),
+                           (const bf16*)sc.data_ptr(), (const bf16*)sh.data_ptr(),
+                           (bf16*)out.data_ptr(), S, D, (float)eps);
+    } else if (xc.scalar_type() == at::kFloat) {
+        hipLaunchKernelGGL(layer_norm_mod_kernel<float>, grid, dim3(256), 0,
+                           cur_stream(), xc.data_ptr<float>(),
+                           sc.data_ptr<float>(), sh.data_ptr<float>(),
+                           out.data_ptr<float>(), S, D, (float)eps);
+    } else {
+        TORCH_CHECK(false, "layer_norm_mod: unsupported dtype");
+    }
+    return out;
+}
+
+at::Tensor gate_residual(at::Tensor res, at::Tensor gate, at::Tensor x) {
+    CHECK_GPU(x);
+    TORCH_CHECK(x.dim() == 3 && gate.dim() == 2, "x [B,S,D], gate [B,D]");
+    auto rc = res.contiguous();
+    auto gc = gate.contiguous();
+    auto xc = x.contiguous();
+    auto out = at::empty_like(xc);
+    const long total = xc.numel();
+    const int S = (int)xc.size(1), D = (int)xc.size(2);
+    const int blocks = (int)std::min<long>((total + 255) / 256, 4096);
+    if (xc.scalar_type() == at::kBFloat16) {
+        hipLaunchKernelGGL(gate_residual_kernel<bf16>, dim3(blocks), dim3(256), 0,
+                           cur_stream(), (const bf16*)rc.data_ptr(),
+                           (const bf16*)gc.data_ptr(), (const bf16*)xc.data_ptr(),
+                           (bf16*)out.data_ptr(), total, S, D);
+    } else if (xc.scalar_type() == at::kFloat) {
+        hipLaunchKernelGGL(gate_residual_kernel<float>, dim3(blocks), dim3(256), 0,
+                           cur_stream(), rc.data_ptr<float>(), gc.data_ptr<float>(),
+                           xc.data_ptr<float>(), out.data_ptr<float>(), total, S, D);
+    } else {
+        TORCH_CHECK(false, "gate_residual: unsupported dtype");
+    }
+    return out;
+}
+
+at::Tensor group_norm_silu(at::Tensor x, long groups,
+                           std::optional<at::Tensor> w,
+                           std::optional<at::Tensor> b, double eps) {
+    CHECK_GPU(x);
+    TORCH_CHECK(x.dim() == 4, "group_norm_silu expects [B, C, H, W]");
+    auto xc = x.contiguous();
+    auto out = at::empty_like(xc);
+    const int B = (int)xc.size(0), C = (int)xc.size(1);
+    const long HW = (long)xc.size(2) * xc.size(3);
+    TORCH_CHECK(C % groups == 0, "C % groups != 0");
+    at::Tensor wf, bf;
+    const float *wp = nullptr, *bp = nullptr;
+    if (w.has_value()) {
+        wf = w->to(at::kFloat).contiguous();
+        bf = b.has_value() ? b->to(at::kFloat).contiguous()
+                           : at::zeros({C}, wf.options());
+        wp = wf.data_ptr<float>();
+        bp = bf.data_ptr<float>();
+    }
+    dim3 grid((unsigned)(B * groups));
+    if (xc.scalar_type() == at::kBFloat16) {
+        hipLaunchKernelGGL(group_norm_silu_kernel<bf16>, grid, dim3(256), 0,
+                           cur_stream(), (const bf16*)xc.data_ptr(), wp, bp,
+                           (bf16*)out.data_ptr(), C, HW, (int)groups, (float)eps);
+    } else if (xc.scalar_type() == at::kFloat) {
+        hipLaunchKernelGGL(group_norm_silu_kernel<float>, grid, dim3(256), 0,
+                           cur_stream(), xc.data_ptr<float>(), wp, bp,
+                           out.data_ptr<float>(), C, HW, (int)groups, (float)eps);
+    } else {
+        TORCH_CHECK(false, "group_norm_silu: unsupported dtype");
+    }
+    return out;
+}
+
+at::Tensor rope_apply(at::Tensor x, at::Tensor cs) {
+    CHECK_GPU(x);
+    TORCH_CHECK(x.dim() == 4, "rope_apply expects [B, H, S, D]");
+    TORCH_CHECK(cs.dim() == 3, "cs expects [S, D/2, 2]");
+    auto xc = x.contiguous();
+    auto cc = cs.to(at::kFloat).contiguous();
+    auto out = at::empty_like(xc);
+    const int S = (int)xc.size(2), D = (int)xc.size(3);
+    TORCH_CHECK((int)cc.size(0) == S && (int)cc.size(1) == D / 2, "cs shape");
+    const long pairs = xc.numel() / 2;
+    const int blocks = (int)std::min<long>((pairs + 255) / 256, 4096);
+    if (xc.scalar_type() == at::kBFloat16) {
+        hipLaunchKernelGGL(rope_apply_kernel<bf16>, dim3(blocks), dim3(256), 0,
+                           cur_stream(), (const bf16*)xc.data_ptr(),
+                           cc.data_ptr<float>(), (bf16*)out.data_ptr(),
+                           pairs, S, D / 2);
+    } else if (xc.scalar_type() == at::kFloat) {
+        hipLaunchKernelGGL(rope_apply_kernel<float>, dim3(blocks), dim3(256), 0,
+                           cur_stream(), xc.data_ptr<float>(),
+                           cc.data_ptr<float>(), out.data_ptr<float>(),
+                           pairs, S, D / 2);
+    } else {
+        TORCH_CHECK(false, "rope_apply: unsupported dtype");
+    }
+    return out;
+}
+
+at::Tensor timestep_embedding(at::Tensor t, long dim, double max_period,
+                              double time_factor) {
+    CHECK_GPU(t);
+    auto tc = t.to(at::kFloat).contiguous();
+    const int B = (int)tc.size(0);
+    auto out = at::empty({B, dim}, tc.options());
+    const int blocks = (int)std::min<long>((B * (dim / 2) + 255) / 256, 1024);
+    hipLaunchKernelGGL(timestep_embedding_kernel, dim3(std::max(blocks, 1)),
+                       dim3(256), 0, cur_stream(), tc.data_ptr<float>(),
+                       out.data_ptr<float>(), B, (int)dim, (float)max_period,
+                       (float)time_factor);
+    return out;
+}
+
+at::Tensor attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v, double scale) {
+    CHECK_GPU(q);
+    TORCH_CHECK(q.scalar_type() == at::kBFloat16, "attn_fwd: bf16 only");
+    TORCH_CHECK(q.dim() == 4, "attn_fwd expects [B, H, S, D]");
+    auto qc = q.contiguous();
+    auto kc = k.contiguous();
+    auto vc = v.contiguous();
+    TORCH_CHECK(qc.sizes() == kc.sizes() && kc.sizes() == vc.sizes(),
+                "q/k/v shape mismatch");
+    const int B = (int)qc.size(0), H = (int)qc.size(1),
+              S = (int)qc.size(2), D = (int)qc.size(3);
+    TORCH_CHECK(D == 64 || D == 128, "attn_fwd: D must be 64 or 128");
+    auto out = at::empty_like(qc);
+    dim3 grid((unsigned)((S + 63) / 64), (unsigned)((long)B * H));
+    if (D == 128) {
+        hipLaunchKernelGGL(attn_fwd_kernel<128>, grid, dim3(256), 0, cur_stream(),
+                           (const bf16*)qc.data_ptr(), (const bf16*)kc.data_ptr(),
+                           (const bf16*)vc.data_ptr(), (bf16*)out.data_ptr(),
+                           S, (float)scale);
+    } else {
+        hipLaunchKernelGGL(attn_fwd_kernel<64>, grid, dim3(256), 0, cur_stream(),
+                           (const bf16*)qc.data_ptr(), (const bf16*)kc.data_ptr(),
+                           (const bf16*)vc.data_ptr(), (bf16*)out.data_ptr(),
+                           S, (float)scale);
+    }
+    return out;
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.def("rms_norm", &rms_norm, "RMSNorm (gfx950)",
+          py::arg("x"), py::arg("weight") = py::none(), py::arg("eps") = 1e-6);
+    m.def("layer_norm_mod", &layer_norm_mod, "AdaLN-modulated LayerNorm (gfx950)");
+    m.def("gate_residual", &gate_residual, "Gated residual add (gfx950)");
+    m.def("group_norm_silu", &group_norm_silu, "GroupNorm+SiLU (gfx950)",
+          py::arg("x"), py::arg("groups"), py::arg("weight") = py::none(),
+          py::arg("bias") = py::none(), py::arg("eps") = 1e-6);
+    m.def("rope_apply", &rope_apply, "RoPE apply (gfx950)");
+    m.def("timestep_embedding", &timestep_embedding, "Sinusoidal timestep embedding");
+    m.def("attn_fwd", &attn_fwd, "Fused flash attention fwd, bf16 MFMA (gfx950)");
+}

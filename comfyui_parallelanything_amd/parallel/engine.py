@@ -101,6 +101,10 @@ class ParallelEngine:
         while idx < len(chain.devices):
             dev = chain.devices[idx]
             force = force_copy_lead and idx == 0
+            if dev in self.replicas:
+                # duplicate device entries (e.g. [cpu, cpu]) share one replica
+                idx += 1
+                continue
             try:
                 replica = replicate_module(model, dev, force_copy=force)
                 self.replicas[dev] = replica

@@ -100,15 +100,16 @@ def test_setup_oom_drops_device(monkeypatch):
 
     def fake_replicate(src, dev, force_copy=False, non_blocking=True):
         calls.append(dev)
-        if len(calls) == 2:  # second device OOMs
+        if dev == "meta":  # second device "OOMs"
             raise torch.cuda.OutOfMemoryError("fake OOM")
         return real(src, dev, force_copy, non_blocking)
 
     monkeypatch.setattr(eng_mod, "replicate_module", fake_replicate)
     m = nn.Linear(3, 3)
-    eng = eng_mod.ParallelEngine(cpu_chain(40, 40, 20), auto_vram_balance=False)
+    chain = DeviceChain(devices=("cpu", "meta", "cpu"), weights=(0.4, 0.4, 0.2))
+    eng = eng_mod.ParallelEngine(chain, auto_vram_balance=False)
     eng.setup(m)
-    assert len(eng.chain.devices) == 2
+    assert eng.chain.devices == ("cpu", "cpu")
     assert eng.chain.weights[0] == pytest.approx(40 / 60)
 
 

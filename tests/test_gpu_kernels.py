@@ -1,0 +1,163 @@
+"""gfx950 HIP kernel numerics vs the fp32 torch reference (ops/reference.py).
+
+Every test builds bf16 (or fp32) inputs, runs the native kernel on the GPU,
+and compares against the reference computed in fp32 on the SAME inputs.
+Run on an MI355X box:  python -m pytest tests -m gpu -x -q
+"""
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+from comfyui_parallelanything_amd import ops
+from comfyui_parallelanything_amd.ops import reference as R
+
+
+@pytest.fixture(scope="module", autouse=True)
+def require_ext():
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    assert ops.hip_available(), (
+        "native _pa_hip extension must be present on a GPU box"
+    )
+
+
+def _cmp(out, ref, rtol, atol, what=""):
+    torch.testing.assert_close(
+        out.float().cpu(), ref.float().cpu(), rtol=rtol, atol=atol, msg=what
+    )
+
+
+@pytest.mark.parametrize("shape", [(2, 64, 3072), (1, 7, 128), (4, 1, 3584)])
+def test_rms_norm_bf16(shape):
+    x = torch.randn(*shape, device="cuda", dtype=torch.bfloat16)
+    w = torch.randn(shape[-1], device="cuda", dtype=torch.bfloat16)
+    out = ops.rms_norm(x, w)
+    ref = R.rms_norm(x.float(), w.float())
+    _cmp(out, ref, 2e-2, 2e-2, "rms_norm")
+
+
+def test_rms_norm_no_weight():
+    x = torch.randn(3, 33, 128, device="cuda", dtype=torch.bfloat16)
+    out = ops.rms_norm(x, None)
+    _cmp(out, R.rms_norm(x.float(), None), 2e-2, 2e-2)
+
+
+def test_rms_norm_fp32():
+    x = torch.randn(5, 257, device="cuda")
+    w = torch.randn(257, device="cuda")
+    out = ops.rms_norm(x, w)
+    _cmp(out, R.rms_norm(x, w), 1e-5, 1e-5)
+
+
+@pytest.mark.parametrize("B,S,D", [(2, 64, 3072), (3, 17, 64), (1, 4608, 128)])
+def test_layer_norm_mod(B, S, D):
+    x = torch.randn(B, S, D, device="cuda", dtype=torch.bfloat16)
+    sc = torch.randn(B, D, device="cuda", dtype=torch.bfloat16)
+    sh = torch.randn(B, D, device="cuda", dtype=torch.bfloat16)
+    out = ops.layer_norm_mod(x, sc, sh)
+    ref = R.layer_norm_mod(x.float(), sc.float(), sh.float())
+    _cmp(out, ref, 2e-2, 5e-2, "layer_norm_mod")
+
+
+def test_gate_residual():
+    r = torch.randn(2, 33, 512, device="cuda", dtype=torch.bfloat16)
+    g = torch.randn(2, 512, device="cuda", dtype=torch.bfloat16)
+    x = torch.randn(2, 33, 512, device="cuda", dtype=torch.bfloat16)
+    out = ops.gate_residual(r, g, x)
+    _cmp(out, R.gate_residual(r.float(), g.float(), x.float()), 2e-2, 2e-2)
+
+
+@pytest.mark.parametrize("B,C,H,W,G", [(2, 320, 32, 32, 32), (1, 64, 8, 8, 8)])
+def test_group_norm_silu(B, C, H, W, G):
+    x = torch.randn(B, C, H, W, device="cuda", dtype=torch.bfloat16)
+    w = torch.randn(C, device="cuda", dtype=torch.bfloat16)
+    b = torch.randn(C, device="cuda", dtype=torch.bfloat16)
+    out = ops.group_norm_silu(x, G, w, b)
+    ref = R.group_norm_silu(x.float(), G, w.float(), b.float())
+    _cmp(out, ref, 2e-2, 5e-2, "group_norm_silu")
+
+
+@pytest.mark.parametrize("B,H,S,D", [(2, 4, 64, 128), (1, 2, 100, 64)])
+def test_rope_apply(B, H, S, D):
+    x = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16)
+    cs = R.rope_freqs(torch.arange(S, device="cuda"), D)
+    out = ops.rope_apply(x, cs)
+    ref = R.rope_apply(x.float(), cs)
+    _cmp(out, ref, 2e-2, 2e-2, "rope_apply")
+
+
+def test_timestep_embedding():
+    t = torch.rand(8, device="cuda")
+    out = ops.timestep_embedding(t, 256)
+    ref = R.timestep_embedding(t.cpu(), 256)
+    _cmp(out, ref, 1e-4, 1e-4, "timestep_embedding")
+
+
+# ---------------- attention ----------------
+
+@pytest.mark.parametrize(
+    "B,H,S,D",
+    [
+        (1, 1, 64, 64),
+        (2, 4, 256, 128),
+        (1, 2, 100, 128),   # tail: S % 32 != 0
+        (1, 2, 37, 64),     # tail: S < KVBLK boundary cases
+        (1, 24, 4608, 128), # FLUX joint-sequence shape
+    ],
+)
+def test_attn_fwd_vs_fp32_reference(B, H, S, D):
+    torch.manual_seed(0)
+    q = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16)
+    out = ops.attention(q, k, v)
+    ref = R.attention(q.float(), k.float(), v.float())
+    # bf16 P quantization + bf16 output: tolerance ~1e-2 abs
+    _cmp(out, ref, 2e-2, 2e-2, f"attn_fwd {B}x{H}x{S}x{D}")
+
+
+def test_attn_fwd_softmax_scale():
+    q = torch.randn(1, 2, 128, 64, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(1, 2, 128, 64, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(1, 2, 128, 64, device="cuda", dtype=torch.bfloat16)
+    out = ops.attention(q, k, v, scale=0.25)
+    ref = R.attention(q.float(), k.float(), v.float(), scale=0.25)
+    _cmp(out, ref, 2e-2, 2e-2)
+
+
+def test_attn_fwd_outlier_rows():
+    """Spiked scores force large rescale steps in the online softmax."""
+    torch.manual_seed(1)
+    q = torch.randn(1, 1, 256, 128, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(1, 1, 256, 128, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(1, 1, 256, 128, device="cuda", dtype=torch.bfloat16)
+    k[0, 0, 200] *= 8.0  # late tile dominates the max
+    out = ops.attention(q, k, v)
+    ref = R.attention(q.float(), k.float(), v.float())
+    _cmp(out, ref, 2e-2, 2e-2, "attn outlier")
+
+
+def test_model_forward_gpu_tiny():
+    from comfyui_parallelanything_amd.models.registry import flux_inputs, make_flux
+
+    m = make_flux(dev="cuda", dtype=torch.bfloat16, tiny=True)
+    x, t, c, kw = flux_inputs(2, dev="cuda", dtype=torch.bfloat16, tiny=True)
+    with torch.no_grad():
+        out = m(x, t, context=c, **kw)
+    assert out.shape == x.shape
+    assert torch.isfinite(out.float()).all()
+
+
+def test_ops_refuse_eager_without_ext(monkeypatch):
+    """On a GPU, a missing extension must raise, not silently fall back."""
+    import comfyui_parallelanything_amd.ops as O
+
+    monkeypatch.setattr(O, "_EXT", None)
+    monkeypatch.setattr(O, "_EXT_TRIED", True)
+    monkeypatch.setattr(O, "_ALLOW_EAGER", False)
+    x = torch.randn(1, 4, 64, device="cuda", dtype=torch.bfloat16)
+    with pytest.raises(RuntimeError, match="Refusing silent eager fallback"):
+        O.rms_norm(x, None)
