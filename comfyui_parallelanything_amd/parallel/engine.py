@@ -188,6 +188,14 @@ class ParallelEngine:
 
     # -- DP path -------------------------------------------------------
     def _data_parallel(self, devices, sizes, batch, x, timesteps, context, kwargs):
+        from ..utils.profiling import trace_range
+
+        with trace_range("pa::dp_step"):
+            return self._data_parallel_impl(
+                devices, sizes, batch, x, timesteps, context, kwargs
+            )
+
+    def _data_parallel_impl(self, devices, sizes, batch, x, timesteps, context, kwargs):
         x_chunks = split_batch(x, sizes)
         t_chunks = split_batch(timesteps, sizes)
         c_chunks = split_batch(context, sizes) if context is not None else None
