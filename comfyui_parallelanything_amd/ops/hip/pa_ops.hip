@@ -128,6 +128,47 @@ __global__ void rms_norm_bf16_kernel(const bf16* __restrict__ x,
     }
 }
 
+// Small-D fast path (qk-norm: D = head_dim 64..256): one WAVE per row, four
+// rows per 256-thread block; pair loads (uint = 2 bf16) when D % 128 == 0.
+__global__ void rms_norm_bf16_row_kernel(const bf16* __restrict__ x,
+                                         const bf16* __restrict__ w,
+                                         bf16* __restrict__ out,
+                                         long rows, int D, float eps) {
+    const int lane = threadIdx.x & 63;
+    const long row = (long)blockIdx.x * 4 + (threadIdx.x >> 6);
+    if (row >= rows) return;
+    const unsigned int* xr =
+        reinterpret_cast<const unsigned int*>(x + row * (long)D);
+    unsigned int* yr = reinterpret_cast<unsigned int*>(out + row * (long)D);
+    const unsigned int* wr = reinterpret_cast<const unsigned int*>(w);
+    const int P = D / 2;  // pairs per row
+    float acc = 0.f;
+    unsigned int vals[4];  // up to D=512
+    const int n = P / 64 + ((lane < (P % 64)) ? 1 : 0);
+    for (int i = 0; i < n; ++i) {
+        unsigned int u = xr[lane + i * 64];
+        vals[i] = u;
+        float a = bf2f(__ushort_as_bfloat16((unsigned short)(u & 0xffff)));
+        float b = bf2f(__ushort_as_bfloat16((unsigned short)(u >> 16)));
+        acc += a * a + b * b;
+    }
+    acc = wave_reduce_sum(acc);
+    const float rrms = rsqrtf(acc / (float)D + eps);
+    for (int i = 0; i < n; ++i) {
+        unsigned int u = vals[i];
+        float a = bf2f(__ushort_as_bfloat16((unsigned short)(u & 0xffff))) * rrms;
+        float b = bf2f(__ushort_as_bfloat16((unsigned short)(u >> 16))) * rrms;
+        if (w != nullptr) {
+            unsigned int uw = wr[lane + i * 64];
+            a *= bf2f(__ushort_as_bfloat16((unsigned short)(uw & 0xffff)));
+            b *= bf2f(__ushort_as_bfloat16((unsigned short)(uw >> 16)));
+        }
+        yr[lane + i * 64] =
+            (unsigned int)__bfloat16_as_ushort(f2bf(a)) |
+            ((unsigned int)__bfloat16_as_ushort(f2bf(b)) << 16);
+    }
+}
+
 // ---------------------------------------------------------------------------
 // AdaLN-modulated LayerNorm: out = LN(x) * (1 + scale[b]) + shift[b].
 // x: [B, S, D]; scale/shift: [B, D]. One block per (b, s) row.
@@ -262,35 +303,51 @@ __global__ void timestep_embedding_kernel(const float* __restrict__ t,
 // ---------------------------------------------------------------------------
 // Fused attention forward (flash-style, non-causal), bf16, D in {64, 128}.
 //
-// Geometry: 256-thread workgroups (4 waves). Each wave owns QBLK=16 query
-// rows; a workgroup covers 64 rows of one (batch, head). K/V stream through
-// LDS in KVBLK=32-key tiles. MFMA v_mfma_f32_16x16x32_bf16 for both QK^T
-// and P*V; online softmax with per-row running (m, l) held across lanes
-// (C-fragment rows live in 16-lane groups; reductions via __shfl_xor<16).
+// v2 geometry: 512-thread workgroups (8 waves). Each wave owns QBLK=32
+// query rows (two 16-row M-tiles); a workgroup covers 256 rows of one
+// (batch, head). K/V stream through LDS in KVBLK=64-key tiles,
+// double-buffered through REGISTERS (T14 async-stage split: issue tile
+// t+1's global loads right after the barrier, write them to LDS at the top
+// of iteration t+1 — HBM latency hides under tile t's 64 MFMAs per wave).
 //
-// LDS layouts (conflict-free by row padding, guide §6 G4):
-//   K tile  : [KVBLK][D + 8]      row stride 272 B (D=128) -> banks disperse
-//   V tile  : [D][KVBLK + 8]      TRANSPOSED at stage time (PV B-fragment
-//                                  wants 8 consecutive keys per lane)
-//   P tile  : per-wave [16][KVBLK + 8]
+// MFMA v_mfma_f32_16x16x32_bf16 for QK^T and P*V; online softmax in the
+// exp2 domain (hardware v_exp_f32 is exp2; saves a VALU mul per element);
+// per-row running (m, l) reduced across the C-fragment's 16-lane groups.
+//
+// LDS layouts (conflict-free by row padding, guide §6 G4; strides chosen so
+// 16 consecutive rows hit 16 distinct bank groups):
+//   K tile : [KVBLK][D + 8]        row stride 272 B (D=128): i*68 dw % 64 distinct
+//   V tile : [D][KVBLK + 8]        TRANSPOSED at stage time (PV B-fragment
+//                                   wants 8 consecutive keys per lane)
+//   P tile : per-wave [32][KVBLK + 8]
 // ---------------------------------------------------------------------------
 using bf16x8 = short8;
+typedef __attribute__((ext_vector_type(4))) unsigned int uint4v;
 
 PA_DEV f32x4 mfma16x16x32(bf16x8 a, bf16x8 b, f32x4 c) {
     return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
 }
 
+#define PA_LOG2E 1.4426950408889634f
+
 template <int D>
-__global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
+__global__ __launch_bounds__(512, 2) void attn_fwd_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ k,
     const bf16* __restrict__ v, bf16* __restrict__ out,
     int S, float scale) {
-    constexpr int KVBLK = 32;
-    constexpr int QBLK = 16;          // per wave
-    constexpr int WAVES = 4;
+    constexpr int KVBLK = 64;
+    constexpr int QBLK = 32;          // per wave
+    constexpr int WAVES = 8;
+    constexpr int THREADS = WAVES * 64;
     constexpr int KPAD = D + 8;       // K row stride (elements)
     constexpr int VPAD = KVBLK + 8;   // V^T row stride
     constexpr int PPAD = KVBLK + 8;
+    constexpr int KK = D / 32;        // MFMA K-steps per QK 16-col tile
+    constexpr int ND = D / 16;        // PV output col tiles
+    constexpr int NKC = KVBLK / 16;   // QK col tiles per KV tile (4)
+    constexpr int NPS = KVBLK / 32;   // PV K-steps per KV tile (2)
+    // per-thread staging share: KVBLK*D elements over THREADS threads
+    constexpr int KVECS = (KVBLK * D) / (8 * THREADS);   // b128 pieces each
 
     __shared__ bf16 k_lds[KVBLK * KPAD];
     __shared__ bf16 v_lds[D * VPAD];
@@ -302,7 +359,6 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
     const int l16 = lane & 15;        // fragment row / col within 16
     const int lg = lane >> 4;         // 16-lane group id (0..3)
 
-    // block -> (bh, qtile); q rows for this wave:
     const long bh = blockIdx.y;
     const int q0 = blockIdx.x * (WAVES * QBLK) + wid * QBLK;
     const long base = bh * (long)S * D;
@@ -312,143 +368,175 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
     const bf16* vp = v + base;
     bf16* op = out + base;
 
-    // ---- load Q fragments (A-layout): lane holds Q[l16][lg*8 + j + 32*kk]
-    constexpr int KK = D / 32;        // MFMA K-steps per QK tile
-    bf16x8 qfrag[KK];
-    {
-        const int row = q0 + l16;
-        const int rr = row < S ? row : S - 1;
+    // ---- Q fragments (A-layout): lane holds Q[m*16 + l16][kk*32 + lg*8 + j]
+    bf16x8 qfrag[2][KK];
 #pragma unroll
-        for (int kk = 0; kk < KK; ++kk) {
-            qfrag[kk] = *reinterpret_cast<const bf16x8*>(
+    for (int m = 0; m < 2; ++m) {
+        const int row = q0 + m * 16 + l16;
+        const int rr = row < S ? row : (S > 0 ? S - 1 : 0);
+#pragma unroll
+        for (int kk = 0; kk < KK; ++kk)
+            qfrag[m][kk] = *reinterpret_cast<const bf16x8*>(
                 qp + (long)rr * D + kk * 32 + lg * 8);
-        }
-        if (row >= S) {
-#pragma unroll
-            for (int kk = 0; kk < KK; ++kk) qfrag[kk] = bf16x8{0,0,0,0,0,0,0,0};
-        }
     }
 
-    // ---- running state: each lane owns 4 rows (r = lg*4 + rr) x cols l16(+16)
-    constexpr int ND = D / 16;        // PV output col tiles
-    f32x4 o_acc[ND];
+    f32x4 o_acc[2][ND];
 #pragma unroll
-    for (int n = 0; n < ND; ++n) o_acc[n] = f32x4{0.f, 0.f, 0.f, 0.f};
-    float m_run[4], l_run[4];
+    for (int m = 0; m < 2; ++m)
 #pragma unroll
-    for (int r = 0; r < 4; ++r) { m_run[r] = -1e30f; l_run[r] = 0.f; }
+        for (int n = 0; n < ND; ++n) o_acc[m][n] = f32x4{0.f, 0.f, 0.f, 0.f};
+    float m_run[2][4], l_run[2][4];
+#pragma unroll
+    for (int m = 0; m < 2; ++m)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) { m_run[m][r] = -1e30f; l_run[m][r] = 0.f; }
 
     bf16* my_p = p_lds + wid * QBLK * PPAD;
+    const float scale2 = scale * PA_LOG2E;   // softmax in exp2 domain
 
-    for (int kv0 = 0; kv0 < S; kv0 += KVBLK) {
-        // ---- stage K tile [KVBLK][D] and V^T tile [D][KVBLK] -------------
-        __syncthreads();
-        {
-            // K: 256 threads load KVBLK*D bf16 with 8-elem vectors
-            constexpr int VECS = KVBLK * D / 8;     // 512 (D=128) / 256 (D=64)
-            for (int i = tid; i < VECS; i += 256) {
-                const int row = i / (D / 8);
-                const int col = (i % (D / 8)) * 8;
-                const int src = kv0 + row;
-                bf16x8 val = (src < S)
-                    ? *reinterpret_cast<const bf16x8*>(kp + (long)src * D + col)
-                    : bf16x8{0,0,0,0,0,0,0,0};
-                *reinterpret_cast<bf16x8*>(&k_lds[row * KPAD + col]) = val;
-                // V transposed: element [row][col+j] -> v_lds[(col+j)][row]
-                bf16x8 vv = (src < S)
-                    ? *reinterpret_cast<const bf16x8*>(vp + (long)src * D + col)
-                    : bf16x8{0,0,0,0,0,0,0,0};
+    // ---- register prefetch state: each thread stages KVECS b128 pieces of
+    //      K and of V. Piece i covers row = (tid + i*THREADS) / (D/8),
+    //      col8 = (tid + i*THREADS) % (D/8).
+    bf16x8 kreg[KVECS], vreg[KVECS];
+
+    auto issue_tile_loads = [&](int kv0) {
 #pragma unroll
-                for (int j = 0; j < 8; ++j)
-                    v_lds[(col + j) * VPAD + row] =
-                        __ushort_as_bfloat16((unsigned short)vv[j]);
+        for (int i = 0; i < KVECS; ++i) {
+            const int idx = tid + i * THREADS;
+            const int row = idx / (D / 8);
+            const int col = (idx % (D / 8)) * 8;
+            const int src = kv0 + row;
+            if (src < S) {
+                kreg[i] = *reinterpret_cast<const bf16x8*>(kp + (long)src * D + col);
+                vreg[i] = *reinterpret_cast<const bf16x8*>(vp + (long)src * D + col);
+            } else {
+                kreg[i] = bf16x8{0,0,0,0,0,0,0,0};
+                vreg[i] = bf16x8{0,0,0,0,0,0,0,0};
             }
         }
-        __syncthreads();
+    };
 
-        // ---- QK^T: 2 col tiles x KK MFMAs ---------------------------------
-        f32x4 s_acc[2];
+    auto write_tile_lds = [&]() {
 #pragma unroll
-        for (int n = 0; n < 2; ++n) {
-            s_acc[n] = f32x4{0.f, 0.f, 0.f, 0.f};
+        for (int i = 0; i < KVECS; ++i) {
+            const int idx = tid + i * THREADS;
+            const int row = idx / (D / 8);
+            const int col = (idx % (D / 8)) * 8;
+            *reinterpret_cast<bf16x8*>(&k_lds[row * KPAD + col]) = kreg[i];
+#pragma unroll
+            for (int j = 0; j < 8; ++j)
+                v_lds[(col + j) * VPAD + row] =
+                    __ushort_as_bfloat16((unsigned short)vreg[i][j]);
+        }
+    };
+
+    issue_tile_loads(0);
+
+    const int n_tiles = (S + KVBLK - 1) / KVBLK;
+    for (int t = 0; t < n_tiles; ++t) {
+        const int kv0 = t * KVBLK;
+        __syncthreads();           // previous tile's LDS reads complete
+        write_tile_lds();
+        __syncthreads();           // tile ready
+        if (t + 1 < n_tiles) issue_tile_loads(kv0 + KVBLK);  // overlap w/ MFMAs
+
+        // ---- QK^T: 2 M-tiles x NKC col tiles x KK K-steps ----------------
+        f32x4 s_acc[2][NKC];
+#pragma unroll
+        for (int n = 0; n < NKC; ++n) {
             const int key = n * 16 + l16;
+            bf16x8 bfrag[KK];
 #pragma unroll
-            for (int kk = 0; kk < KK; ++kk) {
-                bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(
+            for (int kk = 0; kk < KK; ++kk)
+                bfrag[kk] = *reinterpret_cast<const bf16x8*>(
                     &k_lds[key * KPAD + kk * 32 + lg * 8]);
-                s_acc[n] = mfma16x16x32(qfrag[kk], bfrag, s_acc[n]);
+#pragma unroll
+            for (int m = 0; m < 2; ++m) {
+                s_acc[m][n] = f32x4{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+                for (int kk = 0; kk < KK; ++kk)
+                    s_acc[m][n] = mfma16x16x32(qfrag[m][kk], bfrag[kk], s_acc[m][n]);
             }
         }
 
-        // ---- online softmax ----------------------------------------------
-        // s_acc[n][r] = S[q0 + lg*4 + r][kv0 + n*16 + l16] * (pre-scale)
-        // mask tail keys (kv index >= S) to -inf before max/exp
-        const bool key_ok0 = (kv0 + 0 * 16 + l16) < S;
-        const bool key_ok1 = (kv0 + 1 * 16 + l16) < S;
-        float pmax[4];
+        // ---- online softmax (exp2 domain) --------------------------------
+        bool key_ok[NKC];
 #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-            float a = key_ok0 ? s_acc[0][r] * scale : -1e30f;
-            float b = key_ok1 ? s_acc[1][r] * scale : -1e30f;
-            s_acc[0][r] = a; s_acc[1][r] = b;
-            float mx = fmaxf(a, b);
-#pragma unroll
-            for (int off = 1; off < 16; off <<= 1)
-                mx = fmaxf(mx, __shfl_xor(mx, off, 64));
-            pmax[r] = mx;
-        }
-        float alpha[4];
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-            const float mnew = fmaxf(m_run[r], pmax[r]);
-            alpha[r] = __expf(m_run[r] - mnew);
-            m_run[r] = mnew;
-            float p0 = __expf(s_acc[0][r] - mnew);
-            float p1 = __expf(s_acc[1][r] - mnew);
-            s_acc[0][r] = p0; s_acc[1][r] = p1;
-            float ps = p0 + p1;
-#pragma unroll
-            for (int off = 1; off < 16; off <<= 1)
-                ps += __shfl_xor(ps, off, 64);
-            l_run[r] = l_run[r] * alpha[r] + ps;
-        }
-        // rescale O accumulator
-#pragma unroll
-        for (int n = 0; n < ND; ++n)
-#pragma unroll
-            for (int r = 0; r < 4; ++r) o_acc[n][r] *= alpha[r];
+        for (int n = 0; n < NKC; ++n) key_ok[n] = (kv0 + n * 16 + l16) < S;
 
-        // ---- P -> LDS (bf16) then PV --------------------------------------
-        // lane writes P[row = lg*4 + r][key = n*16 + l16]
 #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-            my_p[(lg * 4 + r) * PPAD + l16] = f2bf(s_acc[0][r]);
-            my_p[(lg * 4 + r) * PPAD + 16 + l16] = f2bf(s_acc[1][r]);
+        for (int m = 0; m < 2; ++m) {
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                float sv[NKC];
+                float mx = -1e30f;
+#pragma unroll
+                for (int n = 0; n < NKC; ++n) {
+                    sv[n] = key_ok[n] ? s_acc[m][n][r] * scale2 : -1e30f;
+                    mx = fmaxf(mx, sv[n]);
+                }
+#pragma unroll
+                for (int off = 1; off < 16; off <<= 1)
+                    mx = fmaxf(mx, __shfl_xor(mx, off, 64));
+                const float mnew = fmaxf(m_run[m][r], mx);
+                const float alpha = exp2f(m_run[m][r] - mnew);
+                m_run[m][r] = mnew;
+                float ps = 0.f;
+#pragma unroll
+                for (int n = 0; n < NKC; ++n) {
+                    sv[n] = exp2f(sv[n] - mnew);
+                    ps += sv[n];
+                    s_acc[m][n][r] = sv[n];
+                }
+#pragma unroll
+                for (int off = 1; off < 16; off <<= 1)
+                    ps += __shfl_xor(ps, off, 64);
+                l_run[m][r] = l_run[m][r] * alpha + ps;
+#pragma unroll
+                for (int n = 0; n < ND; ++n) o_acc[m][n][r] *= alpha;
+            }
         }
-        __builtin_amdgcn_s_waitcnt(0);   // lgkm drain (wave-local LDS)
+
+        // ---- P -> LDS (bf16), then PV ------------------------------------
 #pragma unroll
-        for (int n = 0; n < ND; ++n) {
-            // A-frag: P[l16][lg*8 + j]; B-frag: V^T[dim = n*16 + l16][key = lg*8 + j]
-            bf16x8 pa = *reinterpret_cast<const bf16x8*>(
-                &my_p[l16 * PPAD + lg * 8]);
-            bf16x8 vb = *reinterpret_cast<const bf16x8*>(
-                &v_lds[(n * 16 + l16) * VPAD + lg * 8]);
-            o_acc[n] = mfma16x16x32(pa, vb, o_acc[n]);
+        for (int m = 0; m < 2; ++m)
+#pragma unroll
+            for (int n = 0; n < NKC; ++n)
+#pragma unroll
+                for (int r = 0; r < 4; ++r)
+                    my_p[(m * 16 + lg * 4 + r) * PPAD + n * 16 + l16] =
+                        f2bf(s_acc[m][n][r]);
+        // wave-local LDS write->read: drain lgkm (hipcc also tracks these).
+        __builtin_amdgcn_s_waitcnt(/*lgkmcnt(0) vmcnt(max)*/ 0xc07f);
+#pragma unroll
+        for (int ks = 0; ks < NPS; ++ks) {
+#pragma unroll
+            for (int m = 0; m < 2; ++m) {
+                bf16x8 pa = *reinterpret_cast<const bf16x8*>(
+                    &my_p[(m * 16 + l16) * PPAD + ks * 32 + lg * 8]);
+#pragma unroll
+                for (int n = 0; n < ND; ++n) {
+                    bf16x8 vb = *reinterpret_cast<const bf16x8*>(
+                        &v_lds[(n * 16 + l16) * VPAD + ks * 32 + lg * 8]);
+                    o_acc[m][n] = mfma16x16x32(pa, vb, o_acc[m][n]);
+                }
+            }
         }
     }
 
     // ---- epilogue: O / l, store -------------------------------------------
-    // lane holds O[row = q0 + lg*4 + r][dim = n*16 + l16]
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-        const int row = q0 + lg * 4 + r;
-        if (row >= S) continue;
-        const float inv_l = (l_run[r] > 0.f) ? 1.f / l_run[r] : 0.f;
+    for (int m = 0; m < 2; ++m)
 #pragma unroll
-        for (int n = 0; n < ND; ++n) {
-            op[(long)row * D + n * 16 + l16] = f2bf(o_acc[n][r] * inv_l);
+        for (int r = 0; r < 4; ++r) {
+            const int row = q0 + m * 16 + lg * 4 + r;
+            if (row >= S) continue;
+            const float inv_l =
+                (l_run[m][r] > 0.f) ? 1.f / l_run[m][r] : 0.f;
+#pragma unroll
+            for (int n = 0; n < ND; ++n)
+                op[(long)row * D + n * 16 + l16] = f2bf(o_acc[m][n][r] * inv_l);
         }
-    }
 }
 
 // ===========================================================================
@@ -474,7 +562,13 @@ at::Tensor rms_norm(at::Tensor x, std::optional<at::Tensor> w, double eps) {
         wc = w->contiguous();
         TORCH_CHECK(wc.scalar_type() == xc.scalar_type(), "weight dtype mismatch");
     }
-    if (xc.scalar_type() == at::kBFloat16 && (D % 8) == 0) {
+    if (xc.scalar_type() == at::kBFloat16 && (D % 128) == 0 && D <= 512) {
+        hipLaunchKernelGGL(rms_norm_bf16_row_kernel,
+                           dim3((unsigned)((rows + 3) / 4)), dim3(256), 0,
+                           cur_stream(), (const bf16*)xc.data_ptr(),
+                           w.has_value() ? (const bf16*)wc.data_ptr() : nullptr,
+                           (bf16*)out.data_ptr(), rows, D, (float)eps);
+    } else if (xc.scalar_type() == at::kBFloat16 && (D % 8) == 0) {
         hipLaunchKernelGGL(rms_norm_bf16_kernel, dim3((unsigned)rows), dim3(256), 0,
                            cur_stream(),
                            (const bf16*)xc.data_ptr(),
@@ -636,14 +730,14 @@ at::Tensor attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v, double scale) {
               S = (int)qc.size(2), D = (int)qc.size(3);
     TORCH_CHECK(D == 64 || D == 128, "attn_fwd: D must be 64 or 128");
     auto out = at::empty_like(qc);
-    dim3 grid((unsigned)((S + 63) / 64), (unsigned)((long)B * H));
+    dim3 grid((unsigned)((S + 255) / 256), (unsigned)((long)B * H));
     if (D == 128) {
-        hipLaunchKernelGGL(attn_fwd_kernel<128>, grid, dim3(256), 0, cur_stream(),
+        hipLaunchKernelGGL(attn_fwd_kernel<128>, grid, dim3(512), 0, cur_stream(),
                            (const bf16*)qc.data_ptr(), (const bf16*)kc.data_ptr(),
                            (const bf16*)vc.data_ptr(), (bf16*)out.data_ptr(),
                            S, (float)scale);
     } else {
-        hipLaunchKernelGGL(attn_fwd_kernel<64>, grid, dim3(256), 0, cur_stream(),
+        hipLaunchKernelGGL(attn_fwd_kernel<64>, grid, dim3(512), 0, cur_stream(),
                            (const bf16*)qc.data_ptr(), (const bf16*)kc.data_ptr(),
                            (const bf16*)vc.data_ptr(), (bf16*)out.data_ptr(),
                            S, (float)scale);

@@ -38,6 +38,25 @@ def _iter_tensors(module: nn.Module):
         yield b
 
 
+# Device-bound runtime caches that must never be shared across replicas —
+# the reference scrubbed an attr list on every clone (clear_flux_caches,
+# any_device_parallel.py:166-195); our models keep per-device caches in
+# dicts with these names, which are simply reset on the new replica so it
+# repopulates on its own GPU.
+CACHE_ATTRS = ("_pe_cache",)
+
+
+def clear_replica_caches(module: nn.Module) -> int:
+    n = 0
+    for sub in module.modules():
+        for attr in CACHE_ATTRS:
+            cache = getattr(sub, attr, None)
+            if isinstance(cache, dict) and cache:
+                cache.clear()
+                n += 1
+    return n
+
+
 def replicate_module(
     src: nn.Module,
     device,
@@ -70,6 +89,7 @@ def replicate_module(
         memo[id(b)] = data.to(target, non_blocking=non_blocking)
 
     replica = copy.deepcopy(src, memo)
+    clear_replica_caches(replica)
     replica.eval()
     for p in replica.parameters():
         p.requires_grad_(False)

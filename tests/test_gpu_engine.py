@@ -1,0 +1,88 @@
+"""In-process engine tests on a real HIP device (1 GPU is enough).
+
+Covers the ComfyUI-node execution mode: lead-only on GPU, the mixed
+cuda+cpu hybrid chain the reference supports (README.md:133-134 of the
+reference), and pipeline mode with a GPU lead."""
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+from comfyui_parallelanything_amd.models.registry import (
+    flux_inputs,
+    make_flux,
+    make_sd15,
+    sd15_inputs,
+)
+from comfyui_parallelanything_amd.parallel.chain import DeviceChain, make_entry
+from comfyui_parallelanything_amd.parallel.cleanup import cleanup_parallel_model
+from comfyui_parallelanything_amd.parallel.engine import (
+    ParallelEngine,
+    install_parallel_forward,
+)
+from comfyui_parallelanything_amd.parallel.pipeline import configure_pipeline
+
+
+def chain(*entries):
+    return DeviceChain.from_list([make_entry(d, p) for d, p in entries])
+
+
+def test_lead_only_gpu():
+    m = make_flux(dev="cuda:0", dtype=torch.bfloat16, tiny=True)
+    eng = ParallelEngine(chain(("cuda:0", 100)), auto_vram_balance=False)
+    eng.setup(m)
+    x, t, c, kw = flux_inputs(2, dev="cuda:0", dtype=torch.bfloat16, tiny=True)
+    out = eng.forward(x, t, context=c, **kw)
+    assert out.shape == x.shape and out.is_cuda
+    assert torch.isfinite(out.float()).all()
+
+
+def test_hybrid_cuda_cpu_dp_matches_single():
+    """[cuda:0 (50%), cpu (50%)] split == single-device output (fp32 tiny;
+    the cpu replica runs the reference ops, the GPU replica the HIP ops —
+    tolerance covers the kernel-vs-reference numerics)."""
+    m = make_sd15(dev="cuda:0", dtype=torch.float32, tiny=True)
+    x, t, c, kw = sd15_inputs(4, dev="cuda:0", tiny=True)
+    ref = m(x, t, context=c, **kw)
+    eng = ParallelEngine(
+        chain(("cuda:0", 50), ("cpu", 50)), auto_vram_balance=False
+    )
+    eng.setup(m)
+    out = eng.forward(x, t, context=c, **kw)
+    assert out.device.type == "cuda"
+    torch.testing.assert_close(out, ref, rtol=5e-3, atol=5e-3)
+    eng.release()
+
+
+def test_pipeline_mode_gpu_lead():
+    m = make_flux(dev="cuda:0", dtype=torch.float32, tiny=True)
+    x, t, c, kw = flux_inputs(1, dev="cuda:0", dtype=torch.float32, tiny=True)
+    ref = m(x, t, context=c, **kw)
+    eng = ParallelEngine(
+        chain(("cuda:0", 50), ("cpu", 50)), auto_vram_balance=False
+    )
+    eng.setup(m, force_copy_lead=True)
+    configure_pipeline(eng)
+    assert eng.pipeline is not None
+    out = eng.forward(x, t, context=c, **kw)
+    torch.testing.assert_close(out, ref, rtol=5e-3, atol=5e-3)
+    eng.release()
+
+
+def test_install_forward_cleanup_gpu():
+    m = make_flux(dev="cuda:0", dtype=torch.bfloat16, tiny=True)
+    eng = ParallelEngine(chain(("cuda:0", 100)), auto_vram_balance=False)
+    eng.setup(m)
+    install_parallel_forward(m, eng)
+    x, t, c, kw = flux_inputs(2, dev="cuda:0", dtype=torch.bfloat16, tiny=True)
+    out = m(x, t, context=c, **kw)
+    assert out.shape == x.shape
+    cleanup_parallel_model(m)
+    assert not getattr(m, "_true_parallel_active", False)
+
+
+def test_vram_balancer_reads_hbm():
+    from comfyui_parallelanything_amd.parallel.balance import get_free_vram_mb
+
+    free = get_free_vram_mb("cuda:0")
+    assert free > 10_000, f"expected >10 GB free HBM, got {free} MB"
