@@ -1,0 +1,9 @@
+import sys, torch
+sys.path.insert(0, ".")
+from comfyui_parallelanything_amd import ops
+q = torch.randn(8, 24, 4608, 128, device="cuda", dtype=torch.bfloat16)
+k = torch.randn_like(q); v = torch.randn_like(q)
+for _ in range(3): ops.attention(q, k, v)
+torch.cuda.synchronize()
+for _ in range(10): ops.attention(q, k, v)
+torch.cuda.synchronize()
