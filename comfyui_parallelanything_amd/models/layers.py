@@ -120,20 +120,14 @@ def rope_2d_table(
 
 
 class JointAttention(nn.Module):
-    """Attention core shared by MMDiT blocks: qk-norm + rope + fused attn."""
+    """Attention core shared by MMDiT blocks (scale holder; the heavy
+    lifting is ops.attention_bshd on fused-qkv views)."""
 
     def __init__(self, num_heads: int, head_dim: int):
         super().__init__()
         self.num_heads = num_heads
         self.head_dim = head_dim
         self.scale = 1.0 / math.sqrt(head_dim)
-
-    def forward(self, q, k, v, pe: Optional[torch.Tensor]):
-        if pe is not None:
-            q = ops.rope_apply(q, pe)
-            k = ops.rope_apply(k, pe)
-        out = ops.attention(q, k, v, self.scale)
-        return merge_heads(out)
 
 
 class DoubleStreamBlock(nn.Module):
@@ -163,12 +157,15 @@ class DoubleStreamBlock(nn.Module):
         )
         self.attn = JointAttention(num_heads, head_dim)
 
-    def _qkv(self, x, qkv_layer, norm):
-        q, k, v = qkv_layer(x).chunk(3, dim=-1)
-        q = split_heads(q, self.num_heads)
-        k = split_heads(k, self.num_heads)
-        v = split_heads(v, self.num_heads)
-        q, k = norm(q, k)
+    def _qkv(self, x, qkv_layer, norm, pe_slice):
+        """Fused-qkv projection -> strided [B,S,H,D] views (no transposes),
+        with in-place fused qk RMSNorm + RoPE on the projection buffer."""
+        B, S, _ = x.shape
+        qkv = qkv_layer(x).unflatten(-1, (3, self.num_heads, -1))
+        q, k, v = qkv.unbind(2)
+        ops.qk_norm_rope_(
+            q, k, norm.query_norm.scale, norm.key_norm.scale, pe_slice
+        )
         return q, k, v
 
     def forward(self, img, txt, vec, pe):
@@ -177,15 +174,18 @@ class DoubleStreamBlock(nn.Module):
 
         img_in = ops.layer_norm_mod(img, img_m1.scale, img_m1.shift)
         txt_in = ops.layer_norm_mod(txt, txt_m1.scale, txt_m1.shift)
-        iq, ik, iv = self._qkv(img_in, self.img_attn_qkv, self.img_attn_norm)
-        tq, tk, tv = self._qkv(txt_in, self.txt_attn_qkv, self.txt_attn_norm)
+        T = txt.shape[1]
+        iq, ik, iv = self._qkv(img_in, self.img_attn_qkv, self.img_attn_norm,
+                               pe[T:])
+        tq, tk, tv = self._qkv(txt_in, self.txt_attn_qkv, self.txt_attn_norm,
+                               pe[:T])
 
         # joint sequence: txt first, then img (FLUX convention)
-        q = torch.cat([tq, iq], dim=2)
-        k = torch.cat([tk, ik], dim=2)
-        v = torch.cat([tv, iv], dim=2)
-        attn = self.attn(q, k, v, pe)
-        txt_attn, img_attn = attn[:, : txt.shape[1]], attn[:, txt.shape[1]:]
+        q = torch.cat([tq, iq], dim=1)
+        k = torch.cat([tk, ik], dim=1)
+        v = torch.cat([tv, iv], dim=1)
+        attn = ops.attention_bshd(q, k, v, self.attn.scale).flatten(2)
+        txt_attn, img_attn = attn[:, :T], attn[:, T:]
 
         img = ops.gate_residual(img, img_m1.gate, self.img_attn_proj(img_attn))
         img = ops.gate_residual(
@@ -209,25 +209,28 @@ class SingleStreamBlock(nn.Module):
         self.num_heads = num_heads
         self.mlp_dim = int(hidden * mlp_ratio)
         self.linear1 = nn.Linear(hidden, hidden * 3 + self.mlp_dim)
-        self.linear2 = nn.Linear(hidden + self.mlp_dim, hidden)
+        # split output projection: linear2([attn | gelu(mlp)]) as two GEMMs —
+        # avoids materializing a strided last-dim concat every block
+        self.linear2_attn = nn.Linear(hidden, hidden)
+        self.linear2_mlp = nn.Linear(self.mlp_dim, hidden, bias=False)
         self.norm = QKNorm(head_dim)
         self.modulation = Modulation(hidden, double=False)
         self.attn = JointAttention(num_heads, head_dim)
         self.mlp_act = nn.GELU(approximate="tanh")
 
     def forward(self, x, vec, pe):
+        B, S, hidden = x.shape
         mod, _ = self.modulation(vec)
         x_in = ops.layer_norm_mod(x, mod.scale, mod.shift)
-        qkv, mlp_in = self.linear1(x_in).split(
-            [3 * x.shape[-1], self.mlp_dim], dim=-1
+        proj = self.linear1(x_in)
+        qkv = proj[..., : 3 * hidden].unflatten(-1, (3, self.num_heads, -1))
+        mlp_in = proj[..., 3 * hidden:]
+        q, k, v = qkv.unbind(2)  # [B,S,H,D] views into proj
+        ops.qk_norm_rope_(
+            q, k, self.norm.query_norm.scale, self.norm.key_norm.scale, pe
         )
-        q, k, v = qkv.chunk(3, dim=-1)
-        q = split_heads(q, self.num_heads)
-        k = split_heads(k, self.num_heads)
-        v = split_heads(v, self.num_heads)
-        q, k = self.norm(q, k)
-        attn = self.attn(q, k, v, pe)
-        out = self.linear2(torch.cat([attn, self.mlp_act(mlp_in)], dim=-1))
+        attn = ops.attention_bshd(q, k, v, self.attn.scale).flatten(2)
+        out = self.linear2_attn(attn) + self.linear2_mlp(self.mlp_act(mlp_in))
         return ops.gate_residual(x, mod.gate, out)
 
 

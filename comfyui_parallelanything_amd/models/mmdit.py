@@ -116,6 +116,7 @@ class Flux(nn.Module):
             self._pe_cache[key] = pe
         return pe
 
+    @torch.no_grad()
     def forward(self, x, timesteps, context=None, y=None, guidance=None, **kwargs):
         cfg = self.cfg
         B = x.shape[0]
@@ -201,6 +202,7 @@ class ZImage(nn.Module):
             self._pe_cache[key] = pe
         return pe
 
+    @torch.no_grad()
     def forward(self, x, timesteps, context=None, **kwargs):
         cfg = self.cfg
         B, C, H, W = x.shape

@@ -18,7 +18,6 @@ import torch
 from torch import nn
 
 from .. import ops
-from .layers import merge_heads, split_heads
 
 
 class GNSiLU(nn.Module):
@@ -61,10 +60,11 @@ class CrossAttention(nn.Module):
 
     def forward(self, x, context=None):
         ctx = x if context is None else context
-        q = split_heads(self.to_q(x), self.num_heads)
-        k = split_heads(self.to_k(ctx), self.num_heads)
-        v = split_heads(self.to_v(ctx), self.num_heads)
-        return self.to_out(merge_heads(ops.attention(q, k, v, self.scale)))
+        H = self.num_heads
+        q = self.to_q(x).unflatten(-1, (H, -1))
+        k = self.to_k(ctx).unflatten(-1, (H, -1))
+        v = self.to_v(ctx).unflatten(-1, (H, -1))
+        return self.to_out(ops.attention_bshd(q, k, v, self.scale).flatten(2))
 
 
 class GEGLU(nn.Module):
@@ -240,6 +240,7 @@ class SDUNet(nn.Module):
         self.out_norm = GNSiLU(ch)
         self.out_conv = nn.Conv2d(ch, cfg.out_channels, 3, padding=1)
 
+    @torch.no_grad()
     def forward(self, x, timesteps, context=None, y=None, **kwargs):
         cfg = self.cfg
         emb = self.time_embed(

@@ -16,7 +16,7 @@ import torch
 from torch import nn
 
 from .. import ops
-from .layers import MLPEmbedder, QKNorm, merge_heads, split_heads
+from .layers import MLPEmbedder, QKNorm
 
 
 def rope_3d_table(f: int, h: int, w: int, axes_dim: Tuple[int, ...],
@@ -58,21 +58,23 @@ class WanBlock(nn.Module):
         shift1, scale1, gate1, shift2, scale2, gate2 = m
 
         h = ops.layer_norm_mod(x, scale1, shift1)
-        q, k, v = self.self_qkv(h).chunk(3, dim=-1)
-        q = split_heads(q, self.num_heads)
-        k = split_heads(k, self.num_heads)
-        v = split_heads(v, self.num_heads)
-        q, k = self.self_norm(q, k)
-        q = ops.rope_apply(q, pe)
-        k = ops.rope_apply(k, pe)
-        attn = merge_heads(ops.attention(q, k, v, self.scale))
+        qkv = self.self_qkv(h).unflatten(-1, (3, self.num_heads, -1))
+        q, k, v = qkv.unbind(2)  # [B,S,H,D] views
+        ops.qk_norm_rope_(
+            q, k, self.self_norm.query_norm.scale,
+            self.self_norm.key_norm.scale, pe,
+        )
+        attn = ops.attention_bshd(q, k, v, self.scale).flatten(2)
         x = ops.gate_residual(x, gate1, self.self_proj(attn))
 
         h = self.norm_cross(x)
-        q = split_heads(self.cross_q(h), self.num_heads)
-        k = split_heads(self.cross_k(context), self.num_heads)
-        v = split_heads(self.cross_v(context), self.num_heads)
-        x = x + self.cross_proj(merge_heads(ops.attention(q, k, v, self.scale)))
+        H = self.num_heads
+        q = self.cross_q(h).unflatten(-1, (H, -1))
+        k = self.cross_k(context).unflatten(-1, (H, -1))
+        v = self.cross_v(context).unflatten(-1, (H, -1))
+        x = x + self.cross_proj(
+            ops.attention_bshd(q, k, v, self.scale).flatten(2)
+        )
 
         h = ops.layer_norm_mod(x, scale2, shift2)
         return ops.gate_residual(x, gate2, self.ffn(h))
@@ -138,6 +140,7 @@ class WanDiT(nn.Module):
             self._pe_cache[key] = pe
         return pe
 
+    @torch.no_grad()
     def forward(self, x, timesteps, context=None, **kwargs):
         cfg = self.cfg
         B, C, F, H, W = x.shape

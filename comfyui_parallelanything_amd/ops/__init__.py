@@ -175,6 +175,44 @@ def attention(q, k, v, scale: Optional[float] = None) -> torch.Tensor:
     return reference.attention(q, k, v, scale)
 
 
+def attention_bshd(q, k, v, scale: Optional[float] = None) -> torch.Tensor:
+    """Fused attention on [B, S, H, D] (strided views allowed — e.g. slices
+    of a fused qkv projection; no transpose/contiguous copies on GPU)."""
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    if q.is_cuda:
+        ext = _require_ext("attn_fwd_bshd")
+        if ext is not None:
+            D = q.shape[-1]
+            if q.dtype == torch.bfloat16 and D in (64, 128):
+                return ext.attn_fwd_bshd(q, k, v, float(scale))
+            _unsupported("attn_fwd_bshd", f"dtype={q.dtype}, D={D}")
+    out = reference.attention(
+        q.permute(0, 2, 1, 3), k.permute(0, 2, 1, 3), v.permute(0, 2, 1, 3),
+        scale,
+    )
+    return out.permute(0, 2, 1, 3)
+
+
+def qk_norm_rope_(q, k, wq, wk, cs, eps: float = 1e-6):
+    """In-place fused qk RMSNorm + RoPE on [B, S, H, D] views.
+
+    Replaces four bandwidth passes (rms q, rms k, rope q, rope k) with one;
+    per-replica cs tables live on the owning GPU (see models.layers).
+    """
+    if q.is_cuda:
+        ext = _require_ext("qk_norm_rope_")
+        if ext is not None and q.dtype == torch.bfloat16 and q.shape[-1] <= 128:
+            ext.qk_norm_rope_(q, k, wq, wk, cs, eps)
+            return q, k
+        _unsupported("qk_norm_rope_", f"dtype={q.dtype}, D={q.shape[-1]}")
+    qn = reference.rms_norm(q, wq, eps).permute(0, 2, 1, 3)
+    kn = reference.rms_norm(k, wk, eps).permute(0, 2, 1, 3)
+    q.copy_(reference.rope_apply(qn, cs).permute(0, 2, 1, 3))
+    k.copy_(reference.rope_apply(kn, cs).permute(0, 2, 1, 3))
+    return q, k
+
+
 def timestep_embedding(t, dim: int, max_period: float = 10000.0,
                        time_factor: float = 1000.0) -> torch.Tensor:
     if t.is_cuda:

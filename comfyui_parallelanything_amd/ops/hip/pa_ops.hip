@@ -300,6 +300,73 @@ __global__ void timestep_embedding_kernel(const float* __restrict__ t,
     }
 }
 
+
+// ---------------------------------------------------------------------------
+// Fused qk-norm + RoPE (in-place): per (b, s, h) row of q and k,
+// RMSNorm over D with per-head-dim weight, then rotary by cs[s].
+// One wave handles one (b, s, h) row of BOTH q and k — single pass over the
+// qkv projection output (strided [B, S, H, D] views), replacing four
+// separate bandwidth passes (rms q, rms k, rope q, rope k).
+// ---------------------------------------------------------------------------
+__global__ void qk_norm_rope_kernel(bf16* __restrict__ q, bf16* __restrict__ k,
+                                    const bf16* __restrict__ wq,
+                                    const bf16* __restrict__ wk,
+                                    const float* __restrict__ cs,
+                                    int S, int H, int D,
+                                    long q_bs, long q_hs, long q_ss,
+                                    long k_bs, long k_hs, long k_ss,
+                                    long n_rows, float eps) {
+    const int lane = threadIdx.x & 63;
+    const long row = ((long)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+    const int pairs = D / 2;  // <= 64 (D up to 128)
+    if (row >= n_rows) return;
+    const long b = row / ((long)S * H);
+    const long sh = row % ((long)S * H);
+    const int s = (int)(sh / H);
+    const int h = (int)(sh % H);
+    if (lane >= pairs) return;
+
+    const float c = cs[((long)s * pairs + lane) * 2 + 0];
+    const float sn = cs[((long)s * pairs + lane) * 2 + 1];
+    const unsigned int uwq = reinterpret_cast<const unsigned int*>(wq)[lane];
+    const unsigned int uwk = reinterpret_cast<const unsigned int*>(wk)[lane];
+
+    {   // q row
+        unsigned int* p = reinterpret_cast<unsigned int*>(
+            q + b * q_bs + (long)s * q_ss + (long)h * q_hs);
+        const unsigned int u = p[lane];
+        float a0 = bf2f(__ushort_as_bfloat16((unsigned short)(u & 0xffff)));
+        float a1 = bf2f(__ushort_as_bfloat16((unsigned short)(u >> 16)));
+        float ss_ = a0 * a0 + a1 * a1;
+#pragma unroll
+        for (int off = 32; off > 0; off >>= 1) ss_ += __shfl_xor(ss_, off, 64);
+        const float rrms = rsqrtf(ss_ / (float)D + eps);
+        a0 = a0 * rrms * bf2f(__ushort_as_bfloat16((unsigned short)(uwq & 0xffff)));
+        a1 = a1 * rrms * bf2f(__ushort_as_bfloat16((unsigned short)(uwq >> 16)));
+        const float o0 = a0 * c - a1 * sn;
+        const float o1 = a0 * sn + a1 * c;
+        p[lane] = (unsigned int)__bfloat16_as_ushort(f2bf(o0)) |
+                  ((unsigned int)__bfloat16_as_ushort(f2bf(o1)) << 16);
+    }
+    {   // k row
+        unsigned int* p = reinterpret_cast<unsigned int*>(
+            k + b * k_bs + (long)s * k_ss + (long)h * k_hs);
+        const unsigned int u = p[lane];
+        float a0 = bf2f(__ushort_as_bfloat16((unsigned short)(u & 0xffff)));
+        float a1 = bf2f(__ushort_as_bfloat16((unsigned short)(u >> 16)));
+        float ss_ = a0 * a0 + a1 * a1;
+#pragma unroll
+        for (int off = 32; off > 0; off >>= 1) ss_ += __shfl_xor(ss_, off, 64);
+        const float rrms = rsqrtf(ss_ / (float)D + eps);
+        a0 = a0 * rrms * bf2f(__ushort_as_bfloat16((unsigned short)(uwk & 0xffff)));
+        a1 = a1 * rrms * bf2f(__ushort_as_bfloat16((unsigned short)(uwk >> 16)));
+        const float o0 = a0 * c - a1 * sn;
+        const float o1 = a0 * sn + a1 * c;
+        p[lane] = (unsigned int)__bfloat16_as_ushort(f2bf(o0)) |
+                  ((unsigned int)__bfloat16_as_ushort(f2bf(o1)) << 16);
+    }
+}
+
 // ---------------------------------------------------------------------------
 // Fused attention forward (flash-style, non-causal), bf16, D in {64, 128}.
 //
@@ -334,7 +401,11 @@ template <int D>
 __global__ __launch_bounds__(512, 2) void attn_fwd_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ k,
     const bf16* __restrict__ v, bf16* __restrict__ out,
-    int S, float scale) {
+    int S, int Sk, float scale, int H,
+    long q_bs, long q_hs, int q_ss,
+    long k_bs, long k_hs, int k_ss,
+    long v_bs, long v_hs, int v_ss,
+    long o_bs, long o_hs, int o_ss) {
     constexpr int KVBLK = 64;
     constexpr int QBLK = 32;          // per wave
     constexpr int WAVES = 8;
@@ -360,13 +431,14 @@ __global__ __launch_bounds__(512, 2) void attn_fwd_kernel(
     const int lg = lane >> 4;         // 16-lane group id (0..3)
 
     const long bh = blockIdx.y;
+    const long b = bh / H;
+    const int h = (int)(bh % H);
     const int q0 = blockIdx.x * (WAVES * QBLK) + wid * QBLK;
-    const long base = bh * (long)S * D;
 
-    const bf16* qp = q + base;
-    const bf16* kp = k + base;
-    const bf16* vp = v + base;
-    bf16* op = out + base;
+    const bf16* qp = q + b * q_bs + (long)h * q_hs;
+    const bf16* kp = k + b * k_bs + (long)h * k_hs;
+    const bf16* vp = v + b * v_bs + (long)h * v_hs;
+    bf16* op = out + b * o_bs + (long)h * o_hs;
 
     // ---- Q fragments (A-layout): lane holds Q[m*16 + l16][kk*32 + lg*8 + j]
     bf16x8 qfrag[2][KK];
@@ -377,7 +449,7 @@ __global__ __launch_bounds__(512, 2) void attn_fwd_kernel(
 #pragma unroll
         for (int kk = 0; kk < KK; ++kk)
             qfrag[m][kk] = *reinterpret_cast<const bf16x8*>(
-                qp + (long)rr * D + kk * 32 + lg * 8);
+                qp + (long)rr * q_ss + kk * 32 + lg * 8);
     }
 
     f32x4 o_acc[2][ND];
@@ -406,9 +478,9 @@ __global__ __launch_bounds__(512, 2) void attn_fwd_kernel(
             const int row = idx / (D / 8);
             const int col = (idx % (D / 8)) * 8;
             const int src = kv0 + row;
-            if (src < S) {
-                kreg[i] = *reinterpret_cast<const bf16x8*>(kp + (long)src * D + col);
-                vreg[i] = *reinterpret_cast<const bf16x8*>(vp + (long)src * D + col);
+            if (src < Sk) {
+                kreg[i] = *reinterpret_cast<const bf16x8*>(kp + (long)src * k_ss + col);
+                vreg[i] = *reinterpret_cast<const bf16x8*>(vp + (long)src * v_ss + col);
             } else {
                 kreg[i] = bf16x8{0,0,0,0,0,0,0,0};
                 vreg[i] = bf16x8{0,0,0,0,0,0,0,0};
@@ -432,7 +504,7 @@ __global__ __launch_bounds__(512, 2) void attn_fwd_kernel(
 
     issue_tile_loads(0);
 
-    const int n_tiles = (S + KVBLK - 1) / KVBLK;
+    const int n_tiles = (Sk + KVBLK - 1) / KVBLK;
     for (int t = 0; t < n_tiles; ++t) {
         const int kv0 = t * KVBLK;
         __syncthreads();           // previous tile's LDS reads complete
@@ -462,7 +534,7 @@ __global__ __launch_bounds__(512, 2) void attn_fwd_kernel(
         // ---- online softmax (exp2 domain) --------------------------------
         bool key_ok[NKC];
 #pragma unroll
-        for (int n = 0; n < NKC; ++n) key_ok[n] = (kv0 + n * 16 + l16) < S;
+        for (int n = 0; n < NKC; ++n) key_ok[n] = (kv0 + n * 16 + l16) < Sk;
 
 #pragma unroll
         for (int m = 0; m < 2; ++m) {
@@ -535,7 +607,8 @@ __global__ __launch_bounds__(512, 2) void attn_fwd_kernel(
                 (l_run[m][r] > 0.f) ? 1.f / l_run[m][r] : 0.f;
 #pragma unroll
             for (int n = 0; n < ND; ++n)
-                op[(long)row * D + n * 16 + l16] = f2bf(o_acc[m][n][r] * inv_l);
+                op[(long)row * o_ss + n * 16 + l16] =
+                    f2bf(o_acc[m][n][r] * inv_l);
         }
 }
 
@@ -717,32 +790,92 @@ at::Tensor timestep_embedding(at::Tensor t, long dim, double max_period,
     return out;
 }
 
-at::Tensor attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v, double scale) {
+
+void qk_norm_rope_(at::Tensor q, at::Tensor k, at::Tensor wq, at::Tensor wk,
+                   at::Tensor cs, double eps) {
+    CHECK_GPU(q);
+    TORCH_CHECK(q.dim() == 4 && k.dim() == 4, "qk_norm_rope_: [B,S,H,D] views");
+    TORCH_CHECK(q.scalar_type() == at::kBFloat16, "qk_norm_rope_: bf16 only");
+    TORCH_CHECK(q.stride(3) == 1 && k.stride(3) == 1, "last dim contiguous");
+    const int B = (int)q.size(0), S = (int)q.size(1), H = (int)q.size(2),
+              D = (int)q.size(3);
+    TORCH_CHECK(D % 2 == 0 && D <= 128, "qk_norm_rope_: D must be even, <=128");
+    TORCH_CHECK((int)k.size(1) == S && (int)k.size(2) == H, "k shape mismatch");
+    auto csf = cs.to(at::kFloat).contiguous();
+    TORCH_CHECK((int)csf.size(0) == S && (int)csf.size(1) == D / 2, "cs shape");
+    auto wqc = wq.contiguous();
+    auto wkc = wk.contiguous();
+    const long rows = (long)B * S * H;
+    const long blocks = (rows * 64 + 255) / 256;
+    hipLaunchKernelGGL(qk_norm_rope_kernel, dim3((unsigned)blocks), dim3(256), 0,
+                       cur_stream(), (bf16*)q.data_ptr(), (bf16*)k.data_ptr(),
+                       (const bf16*)wqc.data_ptr(), (const bf16*)wkc.data_ptr(),
+                       csf.data_ptr<float>(), S, H, D,
+                       q.stride(0), q.stride(2), q.stride(1),
+                       k.stride(0), k.stride(2), k.stride(1), rows, (float)eps);
+}
+
+struct AttnStrides {
+    long bs, hs;
+    int ss;
+};
+
+static AttnStrides strides_of(const at::Tensor& t, int d_axis_check) {
+    TORCH_CHECK(t.stride(3) == 1, "attn: last dim must be contiguous");
+    return AttnStrides{t.stride(0), t.stride(d_axis_check), (int)0};
+}
+
+static at::Tensor attn_fwd_launch(at::Tensor q, at::Tensor k, at::Tensor v,
+                                  double scale, bool bshd) {
     CHECK_GPU(q);
     TORCH_CHECK(q.scalar_type() == at::kBFloat16, "attn_fwd: bf16 only");
-    TORCH_CHECK(q.dim() == 4, "attn_fwd expects [B, H, S, D]");
-    auto qc = q.contiguous();
-    auto kc = k.contiguous();
-    auto vc = v.contiguous();
-    TORCH_CHECK(qc.sizes() == kc.sizes() && kc.sizes() == vc.sizes(),
-                "q/k/v shape mismatch");
-    const int B = (int)qc.size(0), H = (int)qc.size(1),
-              S = (int)qc.size(2), D = (int)qc.size(3);
+    TORCH_CHECK(q.dim() == 4, "attn_fwd expects 4-D q/k/v");
+    // layouts: bshd = [B, S, H, D] (strided views allowed),
+    //          else  = [B, H, S, D] (must be row-contiguous per head)
+    const int b_ax = 0, s_ax = bshd ? 1 : 2, h_ax = bshd ? 2 : 1, d_ax = 3;
+    const int B = (int)q.size(b_ax), S = (int)q.size(s_ax),
+              H = (int)q.size(h_ax), D = (int)q.size(d_ax);
     TORCH_CHECK(D == 64 || D == 128, "attn_fwd: D must be 64 or 128");
-    auto out = at::empty_like(qc);
+    auto fix = [&](at::Tensor t) {
+        if (t.stride(d_ax) != 1 || (t.stride(s_ax) % 8) != 0 ||
+            (t.stride(h_ax) % 8) != 0)
+            t = t.contiguous();
+        return t;
+    };
+    auto qc = fix(q);
+    auto kc = fix(k);
+    auto vc = fix(v);
+    const int Sk = (int)kc.size(s_ax);
+    TORCH_CHECK((int)vc.size(s_ax) == Sk && (int)kc.size(h_ax) == H,
+                "attn: k/v shape mismatch");
+    at::Tensor out = bshd
+        ? at::empty({B, S, H, D}, qc.options())
+        : at::empty({B, H, S, D}, qc.options());
     dim3 grid((unsigned)((S + 255) / 256), (unsigned)((long)B * H));
+#define PA_ATTN_ARGS                                                          \
+    (const bf16*)qc.data_ptr(), (const bf16*)kc.data_ptr(),                   \
+    (const bf16*)vc.data_ptr(), (bf16*)out.data_ptr(), S, Sk, (float)scale, H, \
+    qc.stride(b_ax), qc.stride(h_ax), (int)qc.stride(s_ax),                   \
+    kc.stride(b_ax), kc.stride(h_ax), (int)kc.stride(s_ax),                   \
+    vc.stride(b_ax), vc.stride(h_ax), (int)vc.stride(s_ax),                   \
+    out.stride(b_ax), out.stride(h_ax), (int)out.stride(s_ax)
     if (D == 128) {
-        hipLaunchKernelGGL(attn_fwd_kernel<128>, grid, dim3(512), 0, cur_stream(),
-                           (const bf16*)qc.data_ptr(), (const bf16*)kc.data_ptr(),
-                           (const bf16*)vc.data_ptr(), (bf16*)out.data_ptr(),
-                           S, (float)scale);
+        hipLaunchKernelGGL(attn_fwd_kernel<128>, grid, dim3(512), 0,
+                           cur_stream(), PA_ATTN_ARGS);
     } else {
-        hipLaunchKernelGGL(attn_fwd_kernel<64>, grid, dim3(512), 0, cur_stream(),
-                           (const bf16*)qc.data_ptr(), (const bf16*)kc.data_ptr(),
-                           (const bf16*)vc.data_ptr(), (bf16*)out.data_ptr(),
-                           S, (float)scale);
+        hipLaunchKernelGGL(attn_fwd_kernel<64>, grid, dim3(512), 0,
+                           cur_stream(), PA_ATTN_ARGS);
     }
+#undef PA_ATTN_ARGS
     return out;
+}
+
+at::Tensor attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v, double scale) {
+    return attn_fwd_launch(q, k, v, scale, /*bshd=*/false);
+}
+
+at::Tensor attn_fwd_bshd(at::Tensor q, at::Tensor k, at::Tensor v, double scale) {
+    return attn_fwd_launch(q, k, v, scale, /*bshd=*/true);
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -756,4 +889,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("rope_apply", &rope_apply, "RoPE apply (gfx950)");
     m.def("timestep_embedding", &timestep_embedding, "Sinusoidal timestep embedding");
     m.def("attn_fwd", &attn_fwd, "Fused flash attention fwd, bf16 MFMA (gfx950)");
+    m.def("attn_fwd_bshd", &attn_fwd_bshd,
+          "Fused flash attention fwd on [B,S,H,D] strided views (gfx950)");
+    m.def("qk_norm_rope_", &qk_norm_rope_,
+          "In-place fused qk RMSNorm + RoPE on [B,S,H,D] views (gfx950)");
 }

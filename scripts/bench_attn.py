@@ -9,7 +9,7 @@ import time
 
 import torch
 
-sys.path.insert(0, ".")
+sys.path.insert(0, __import__("os").path.dirname(__import__("os").path.dirname(__import__("os").path.abspath(__file__))))
 from comfyui_parallelanything_amd import ops  # noqa: E402
 
 

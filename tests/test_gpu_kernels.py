@@ -140,6 +140,51 @@ def test_attn_fwd_outlier_rows():
     _cmp(out, ref, 2e-2, 2e-2, "attn outlier")
 
 
+def test_attn_fwd_bshd_strided_views():
+    """BSHD path on strided fused-qkv views == reference on dense copies."""
+    torch.manual_seed(3)
+    B, S, H, D = 2, 200, 4, 128
+    qkv = torch.randn(B, S, 3, H, D, device="cuda", dtype=torch.bfloat16)
+    q, k, v = qkv.unbind(2)  # strided views
+    out = ops.attention_bshd(q, k, v)
+    ref = R.attention(
+        q.permute(0, 2, 1, 3).float(),
+        k.permute(0, 2, 1, 3).float(),
+        v.permute(0, 2, 1, 3).float(),
+    ).permute(0, 2, 1, 3)
+    _cmp(out, ref, 2e-2, 2e-2, "attn bshd strided")
+
+
+def test_attn_fwd_cross_lengths():
+    """Sq != Skv (cross-attention, e.g. 77 text tokens)."""
+    torch.manual_seed(4)
+    q = torch.randn(2, 4, 256, 64, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(2, 4, 77, 64, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(2, 4, 77, 64, device="cuda", dtype=torch.bfloat16)
+    out = ops.attention_bshd(
+        q.permute(0, 2, 1, 3), k.permute(0, 2, 1, 3), v.permute(0, 2, 1, 3)
+    ).permute(0, 2, 1, 3)
+    ref = R.attention(q.float(), k.float(), v.float())
+    _cmp(out, ref, 2e-2, 2e-2, "attn cross")
+
+
+def test_qk_norm_rope_fused():
+    torch.manual_seed(5)
+    B, S, H, D = 2, 100, 4, 128
+    qkv = torch.randn(B, S, 3, H, D, device="cuda", dtype=torch.bfloat16)
+    q, k, v = qkv.unbind(2)
+    q_ref = q.clone().permute(0, 2, 1, 3).float()
+    k_ref = k.clone().permute(0, 2, 1, 3).float()
+    wq = torch.randn(D, device="cuda", dtype=torch.bfloat16)
+    wk = torch.randn(D, device="cuda", dtype=torch.bfloat16)
+    cs = R.rope_freqs(torch.arange(S, device="cuda"), D)
+    ops.qk_norm_rope_(q, k, wq, wk, cs)
+    ref_q = R.rope_apply(R.rms_norm(q_ref, wq.float()), cs)
+    ref_k = R.rope_apply(R.rms_norm(k_ref, wk.float()), cs)
+    _cmp(q.permute(0, 2, 1, 3), ref_q, 3e-2, 3e-2, "fused qk q")
+    _cmp(k.permute(0, 2, 1, 3), ref_k, 3e-2, 3e-2, "fused qk k")
+
+
 def test_model_forward_gpu_tiny():
     from comfyui_parallelanything_amd.models.registry import flux_inputs, make_flux
 
