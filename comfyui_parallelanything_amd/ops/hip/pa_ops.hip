@@ -496,9 +496,15 @@ __global__ __launch_bounds__(512, 2) void attn_fwd_kernel(
             const int col = (idx % (D / 8)) * 8;
             *reinterpret_cast<bf16x8*>(&k_lds[row * KPAD + col]) = kreg[i];
 #pragma unroll
-            for (int j = 0; j < 8; ++j)
-                v_lds[(col + j) * VPAD + row] =
+            for (int j = 0; j < 8; ++j) {
+                const int dim = col + j;
+                // granule-XOR swizzle: keys permuted in 8-wide granules by
+                // (dim>>3)&7; read side applies the same XOR (G4 / T2).
+                const int key_swz =
+                    (((row >> 3) ^ ((dim >> 3) & 7)) << 3) | (row & 7);
+                v_lds[dim * VPAD + key_swz] =
                     __ushort_as_bfloat16((unsigned short)vreg[i][j]);
+            }
         }
     };
 
@@ -522,6 +528,7 @@ __global__ __launch_bounds__(512, 2) void attn_fwd_kernel(
             for (int kk = 0; kk < KK; ++kk)
                 bfrag[kk] = *reinterpret_cast<const bf16x8*>(
                     &k_lds[key * KPAD + kk * 32 + lg * 8]);
+            __builtin_amdgcn_s_setprio(1);
 #pragma unroll
             for (int m = 0; m < 2; ++m) {
                 s_acc[m][n] = f32x4{0.f, 0.f, 0.f, 0.f};
@@ -529,6 +536,7 @@ __global__ __launch_bounds__(512, 2) void attn_fwd_kernel(
                 for (int kk = 0; kk < KK; ++kk)
                     s_acc[m][n] = mfma16x16x32(qfrag[m][kk], bfrag[kk], s_acc[m][n]);
             }
+            __builtin_amdgcn_s_setprio(0);
         }
 
         // ---- online softmax (exp2 domain) --------------------------------
@@ -588,8 +596,10 @@ __global__ __launch_bounds__(512, 2) void attn_fwd_kernel(
                     &my_p[(m * 16 + l16) * PPAD + ks * 32 + lg * 8]);
 #pragma unroll
                 for (int n = 0; n < ND; ++n) {
+                    const int dim = n * 16 + l16;
+                    const int gsw = ((ks * 4 + lg) ^ ((dim >> 3) & 7)) << 3;
                     bf16x8 vb = *reinterpret_cast<const bf16x8*>(
-                        &v_lds[(n * 16 + l16) * VPAD + ks * 32 + lg * 8]);
+                        &v_lds[dim * VPAD + gsw]);
                     o_acc[m][n] = mfma16x16x32(pa, vb, o_acc[m][n]);
                 }
             }
