@@ -219,8 +219,12 @@ class ParallelEngine:
             try:
                 if stream is None:  # cpu worker: synchronous
                     out = self._run_chunk(
-                        dev, x_chunks[i], t_chunks[i],
-                        c_chunks[i] if c_chunks else None, kw_chunks[i]
+                        dev,
+                        move_to_device(x_chunks[i], dev),
+                        move_to_device(t_chunks[i], dev),
+                        move_to_device(c_chunks[i], dev) if c_chunks else None,
+                        {k: move_to_device(v, dev)
+                         for k, v in kw_chunks[i].items()},
                     )
                     results[i] = move_to_device(out, lead)
                     continue
