@@ -420,8 +420,8 @@ __global__ __launch_bounds__(512, 2) void attn_fwd_kernel(
     // per-thread staging share: KVBLK*D elements over THREADS threads
     constexpr int KVECS = (KVBLK * D) / (8 * THREADS);   // b128 pieces each
 
-    __shared__ bf16 k_lds[2][KVBLK * KPAD];
-    __shared__ bf16 v_lds[2][D * VPAD];
+    __shared__ bf16 k_lds[KVBLK * KPAD];
+    __shared__ bf16 v_lds[D * VPAD];
     __shared__ bf16 p_lds[WAVES * QBLK * PPAD];
 
     const int tid = threadIdx.x;
@@ -488,13 +488,13 @@ __global__ __launch_bounds__(512, 2) void attn_fwd_kernel(
         }
     };
 
-    auto write_tile_lds = [&](int buf) {
+    auto write_tile_lds = [&]() {
 #pragma unroll
         for (int i = 0; i < KVECS; ++i) {
             const int idx = tid + i * THREADS;
             const int row = idx / (D / 8);
             const int col = (idx % (D / 8)) * 8;
-            *reinterpret_cast<bf16x8*>(&k_lds[buf][row * KPAD + col]) = kreg[i];
+            *reinterpret_cast<bf16x8*>(&k_lds[row * KPAD + col]) = kreg[i];
 #pragma unroll
             for (int j = 0; j < 8; ++j) {
                 const int dim = col + j;
@@ -502,27 +502,21 @@ __global__ __launch_bounds__(512, 2) void attn_fwd_kernel(
                 // (dim>>3)&7; read side applies the same XOR (G4 / T2).
                 const int key_swz =
                     (((row >> 3) ^ ((dim >> 3) & 7)) << 3) | (row & 7);
-                v_lds[buf][dim * VPAD + key_swz] =
+                v_lds[dim * VPAD + key_swz] =
                     __ushort_as_bfloat16((unsigned short)vreg[i][j]);
             }
         }
     };
 
-    // double-buffered LDS (ONE barrier per tile): regs for tile t+1 are
-    // written into buf (t+1)&1 while tile t computes from buf t&1; the
-    // iteration-top barrier jointly proves "buf[t] writes done" and
-    // "compute(t-1), the previous reader of buf[(t+1)&1], done".
-    const int n_tiles = (Sk + KVBLK - 1) / KVBLK;
     issue_tile_loads(0);
-    write_tile_lds(0);
-    if (n_tiles > 1) issue_tile_loads(KVBLK);
+
+    const int n_tiles = (Sk + KVBLK - 1) / KVBLK;
     for (int t = 0; t < n_tiles; ++t) {
         const int kv0 = t * KVBLK;
-        __syncthreads();  // buf[t&1] ready; buf[(t+1)&1] free for writing
-        if (t + 1 < n_tiles) write_tile_lds((t + 1) & 1);
-        if (t + 2 < n_tiles) issue_tile_loads(kv0 + 2 * KVBLK);
-        const bf16* kb = k_lds[t & 1];
-        const bf16* vb_base = v_lds[t & 1];
+        __syncthreads();           // previous tile's LDS reads complete
+        write_tile_lds();
+        __syncthreads();           // tile ready
+        if (t + 1 < n_tiles) issue_tile_loads(kv0 + KVBLK);  // overlap w/ MFMAs
 
         // ---- QK^T: 2 M-tiles x NKC col tiles x KK K-steps ----------------
         f32x4 s_acc[2][NKC];
@@ -533,7 +527,7 @@ __global__ __launch_bounds__(512, 2) void attn_fwd_kernel(
 #pragma unroll
             for (int kk = 0; kk < KK; ++kk)
                 bfrag[kk] = *reinterpret_cast<const bf16x8*>(
-                    &kb[key * KPAD + kk * 32 + lg * 8]);
+                    &k_lds[key * KPAD + kk * 32 + lg * 8]);
             __builtin_amdgcn_s_setprio(1);
 #pragma unroll
             for (int m = 0; m < 2; ++m) {
@@ -605,7 +599,7 @@ __global__ __launch_bounds__(512, 2) void attn_fwd_kernel(
                     const int dim = n * 16 + l16;
                     const int gsw = ((ks * 4 + lg) ^ ((dim >> 3) & 7)) << 3;
                     bf16x8 vb = *reinterpret_cast<const bf16x8*>(
-                        &vb_base[dim * VPAD + gsw]);
+                        &v_lds[dim * VPAD + gsw]);
                     o_acc[m][n] = mfma16x16x32(pa, vb, o_acc[m][n]);
                 }
             }
