@@ -221,6 +221,75 @@ __global__ void gate_residual_kernel(const T* __restrict__ res,
     }
 }
 
+
+// bf16 fast path: short8 vector loads both passes (mean/var re-read hits L2)
+__global__ void layer_norm_mod_bf16_kernel(const bf16* __restrict__ x,
+                                           const bf16* __restrict__ scale,
+                                           const bf16* __restrict__ shift,
+                                           bf16* __restrict__ out,
+                                           int S, int D, float eps) {
+    const long row = blockIdx.x;
+    const long b = row / S;
+    const short8* xr = reinterpret_cast<const short8*>(x + row * (long)D);
+    const short8* sc = reinterpret_cast<const short8*>(scale + b * (long)D);
+    const short8* sh = reinterpret_cast<const short8*>(shift + b * (long)D);
+    short8* yr = reinterpret_cast<short8*>(out + row * (long)D);
+    const int DV = D / 8;
+    __shared__ float scratch[8];
+
+    float s1 = 0.f, s2 = 0.f;
+    for (int i = threadIdx.x; i < DV; i += blockDim.x) {
+        short8 v = xr[i];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            float f = bf2f(__ushort_as_bfloat16((unsigned short)v[j]));
+            s1 += f;
+            s2 += f * f;
+        }
+    }
+    float mean = block_reduce_sum<256>(s1, scratch) / (float)D;
+    float var = block_reduce_sum<256>(s2, scratch) / (float)D - mean * mean;
+    const float rstd = rsqrtf(var + eps);
+    for (int i = threadIdx.x; i < DV; i += blockDim.x) {
+        short8 v = xr[i], a = sc[i], c = sh[i], o;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            float f = (bf2f(__ushort_as_bfloat16((unsigned short)v[j])) - mean) * rstd;
+            f = f * (1.f + bf2f(__ushort_as_bfloat16((unsigned short)a[j]))) +
+                bf2f(__ushort_as_bfloat16((unsigned short)c[j]));
+            o[j] = (short)__bfloat16_as_ushort(f2bf(f));
+        }
+        yr[i] = o;
+    }
+}
+
+// bf16 fast path: 8-wide gated residual
+__global__ void gate_residual_bf16_kernel(const bf16* __restrict__ res,
+                                          const bf16* __restrict__ gate,
+                                          const bf16* __restrict__ x,
+                                          bf16* __restrict__ out,
+                                          long total8, int S, int DV) {
+    const long stride = (long)gridDim.x * blockDim.x;
+    const short8* rv = reinterpret_cast<const short8*>(res);
+    const short8* gv = reinterpret_cast<const short8*>(gate);
+    const short8* xv = reinterpret_cast<const short8*>(x);
+    short8* ov = reinterpret_cast<short8*>(out);
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total8;
+         i += stride) {
+        const long dv = i % DV;
+        const long b = i / ((long)S * DV);
+        short8 r = rv[i], g = gv[b * DV + dv], xx = xv[i], o;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            float f = bf2f(__ushort_as_bfloat16((unsigned short)r[j])) +
+                      bf2f(__ushort_as_bfloat16((unsigned short)g[j])) *
+                          bf2f(__ushort_as_bfloat16((unsigned short)xx[j]));
+            o[j] = (short)__bfloat16_as_ushort(f2bf(f));
+        }
+        ov[i] = o;
+    }
+}
+
 // ---------------------------------------------------------------------------
 // GroupNorm + SiLU: x [B, C, H, W]; one block per (b, group).
 // ---------------------------------------------------------------------------
@@ -684,7 +753,12 @@ at::Tensor layer_norm_mod(at::Tensor x, at::Tensor scale, at::Tensor shift,
     auto out = at::empty_like(xc);
     const int B = (int)xc.size(0), S = (int)xc.size(1), D = (int)xc.size(2);
     dim3 grid((unsigned)((long)B * S));
-    if (xc.scalar_type() == at::kBFloat16) {
+    if (xc.scalar_type() == at::kBFloat16 && (D % 8) == 0) {
+        hipLaunchKernelGGL(layer_norm_mod_bf16_kernel, grid, dim3(256), 0,
+                           cur_stream(), (const bf16*)xc.data_ptr(),
+                           (const bf16*)sc.data_ptr(), (const bf16*)sh.data_ptr(),
+                           (bf16*)out.data_ptr(), S, D, (float)eps);
+    } else if (xc.scalar_type() == at::kBFloat16) {
         hipLaunchKernelGGL(layer_norm_mod_kernel<bf16>, grid, dim3(256), 0,
                            cur_stream(), (const bf16*)xc.data_ptr(),
                            (const bf16*)sc.data_ptr(), (const bf16*)sh.data_ptr(),
@@ -710,7 +784,12 @@ at::Tensor gate_residual(at::Tensor res, at::Tensor gate, at::Tensor x) {
     const long total = xc.numel();
     const int S = (int)xc.size(1), D = (int)xc.size(2);
     const int blocks = (int)std::min<long>((total + 255) / 256, 4096);
-    if (xc.scalar_type() == at::kBFloat16) {
+    if (xc.scalar_type() == at::kBFloat16 && (D % 8) == 0) {
+        hipLaunchKernelGGL(gate_residual_bf16_kernel, dim3(blocks), dim3(256), 0,
+                           cur_stream(), (const bf16*)rc.data_ptr(),
+                           (const bf16*)gc.data_ptr(), (const bf16*)xc.data_ptr(),
+                           (bf16*)out.data_ptr(), total / 8, S, D / 8);
+    } else if (xc.scalar_type() == at::kBFloat16) {
         hipLaunchKernelGGL(gate_residual_kernel<bf16>, dim3(blocks), dim3(256), 0,
                            cur_stream(), (const bf16*)rc.data_ptr(),
                            (const bf16*)gc.data_ptr(), (const bf16*)xc.data_ptr(),
