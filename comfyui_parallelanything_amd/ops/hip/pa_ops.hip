@@ -608,36 +608,40 @@ __global__ __launch_bounds__(512, 2) void attn_fwd_kernel(
             __builtin_amdgcn_s_setprio(0);
         }
 
-        // ---- per-M-tile softmax -> P stage -> PV --------------------------
-        // softmax(m=1)'s VALU work is independent of PV(m=0)'s MFMAs, so the
-        // scheduler overlaps the two pipes (T15 motif); hipcc tracks the
-        // p_lds write->read dependency with counted lgkm waits.
+        // ---- online softmax (exp2 domain) --------------------------------
         bool key_ok[NKC];
 #pragma unroll
         for (int n = 0; n < NKC; ++n) key_ok[n] = (kv0 + n * 16 + l16) < Sk;
 
+        bool all_skip = true;
+        float mx_mr[2][4];
 #pragma unroll
         for (int m = 0; m < 2; ++m) {
-            bool all_skip = true;
-            float mx_r[4];
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
                 float mx = -1e30f;
 #pragma unroll
                 for (int n = 0; n < NKC; ++n) {
-                    float sv = key_ok[n] ? s_acc[m][n][r] * scale2 : -1e30f;
+                    const float sv = key_ok[n] ? s_acc[m][n][r] * scale2 : -1e30f;
                     s_acc[m][n][r] = sv;
                     mx = fmaxf(mx, sv);
                 }
 #pragma unroll
                 for (int off = 1; off < 16; off <<= 1)
                     mx = fmaxf(mx, __shfl_xor(mx, off, 64));
-                mx_r[r] = mx;
+                mx_mr[m][r] = mx;
                 all_skip = all_skip && (mx <= m_run[m][r]);
             }
+        }
+        // exact defer: when NO row's max grew (common on late tiles),
+        // alpha == exp2(0) == 1 exactly -> skip the O-accumulator rescale
+        const bool skip = __all(all_skip);
+#pragma unroll
+        for (int m = 0; m < 2; ++m) {
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
-                const float mnew = fmaxf(m_run[m][r], mx_r[r]);
+                const float mnew = skip ? m_run[m][r]
+                                        : fmaxf(m_run[m][r], mx_mr[m][r]);
                 float ps = 0.f;
 #pragma unroll
                 for (int n = 0; n < NKC; ++n) {
@@ -648,31 +652,35 @@ __global__ __launch_bounds__(512, 2) void attn_fwd_kernel(
 #pragma unroll
                 for (int off = 1; off < 16; off <<= 1)
                     ps += __shfl_xor(ps, off, 64);
-                if (!__all(all_skip)) {
-                    // at least one row's max grew: exact rescale (alpha may
-                    // be 1 for rows whose max held; exp2(0) == 1 exactly)
+                if (!skip) {
                     const float alpha = exp2f(m_run[m][r] - mnew);
                     m_run[m][r] = mnew;
                     l_run[m][r] = l_run[m][r] * alpha + ps;
 #pragma unroll
                     for (int n = 0; n < ND; ++n) o_acc[m][n][r] *= alpha;
                 } else {
-                    l_run[m][r] += ps;  // every row's max held: alpha == 1
+                    l_run[m][r] += ps;
                 }
             }
-            // P(m) -> LDS (bf16)
+        }
+
+        // ---- P -> LDS (bf16), then PV ------------------------------------
+#pragma unroll
+        for (int m = 0; m < 2; ++m)
 #pragma unroll
             for (int n = 0; n < NKC; ++n)
 #pragma unroll
                 for (int r = 0; r < 4; ++r)
                     my_p[(m * 16 + lg * 4 + r) * PPAD + n * 16 + l16] =
                         f2bf(s_acc[m][n][r]);
-            // PV(m)
+        // wave-local LDS write->read: drain lgkm (hipcc also tracks these).
+        __builtin_amdgcn_s_waitcnt(/*lgkmcnt(0) vmcnt(max)*/ 0xc07f);
 #pragma unroll
-            for (int ks = 0; ks < NPS; ++ks) {
+        for (int ks = 0; ks < NPS; ++ks) {
+#pragma unroll
+            for (int m = 0; m < 2; ++m) {
                 bf16x8 pa = *reinterpret_cast<const bf16x8*>(
                     &my_p[(m * 16 + l16) * PPAD + ks * 32 + lg * 8]);
-                __builtin_amdgcn_s_setprio(1);
 #pragma unroll
                 for (int n = 0; n < ND; ++n) {
                     const int dim = n * 16 + l16;
@@ -681,7 +689,6 @@ __global__ __launch_bounds__(512, 2) void attn_fwd_kernel(
                         &v_lds[dim * VPAD + gsw]);
                     o_acc[m][n] = mfma16x16x32(pa, vb, o_acc[m][n]);
                 }
-                __builtin_amdgcn_s_setprio(0);
             }
         }
     }
