@@ -177,7 +177,12 @@ def attention(q, k, v, scale: Optional[float] = None) -> torch.Tensor:
 
 def attention_bshd(q, k, v, scale: Optional[float] = None) -> torch.Tensor:
     """Fused attention on [B, S, H, D] (strided views allowed — e.g. slices
-    of a fused qkv projection; no transpose/contiguous copies on GPU)."""
+    of a fused qkv projection; no transpose/contiguous copies on GPU).
+
+    Head dims that are not 64/128 but fit under them (SD1.5's 40/80) are
+    zero-padded: padding q/k leaves every softmax score unchanged, padded v
+    columns are sliced off the output.
+    """
     if scale is None:
         scale = 1.0 / math.sqrt(q.shape[-1])
     if q.is_cuda:
@@ -186,6 +191,16 @@ def attention_bshd(q, k, v, scale: Optional[float] = None) -> torch.Tensor:
             D = q.shape[-1]
             if q.dtype == torch.bfloat16 and D in (64, 128):
                 return ext.attn_fwd_bshd(q, k, v, float(scale))
+            if q.dtype == torch.bfloat16 and D < 128:
+                Dp = 64 if D <= 64 else 128
+                pad = (0, Dp - D)
+                out = ext.attn_fwd_bshd(
+                    torch.nn.functional.pad(q, pad),
+                    torch.nn.functional.pad(k, pad),
+                    torch.nn.functional.pad(v, pad),
+                    float(scale),
+                )
+                return out[..., :D]
             _unsupported("attn_fwd_bshd", f"dtype={q.dtype}, D={D}")
     out = reference.attention(
         q.permute(0, 2, 1, 3), k.permute(0, 2, 1, 3), v.permute(0, 2, 1, 3),

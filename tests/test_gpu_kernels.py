@@ -206,3 +206,17 @@ def test_ops_refuse_eager_without_ext(monkeypatch):
     x = torch.randn(1, 4, 64, device="cuda", dtype=torch.bfloat16)
     with pytest.raises(RuntimeError, match="Refusing silent eager fallback"):
         O.rms_norm(x, None)
+
+
+def test_attn_pad_head_dims():
+    """SD1.5-style head dims (40/80) via zero-padding == reference."""
+    for D in (40, 80):
+        torch.manual_seed(D)
+        q = torch.randn(1, 3, 128, D, device="cuda", dtype=torch.bfloat16)
+        k = torch.randn_like(q)
+        v = torch.randn_like(q)
+        out = ops.attention_bshd(
+            q.permute(0, 2, 1, 3), k.permute(0, 2, 1, 3), v.permute(0, 2, 1, 3)
+        ).permute(0, 2, 1, 3)
+        ref = R.attention(q.float(), k.float(), v.float())
+        _cmp(out, ref, 2e-2, 2e-2, f"attn pad D={D}")
