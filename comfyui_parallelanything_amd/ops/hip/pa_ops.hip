@@ -922,6 +922,248 @@ void qk_norm_rope_(at::Tensor q, at::Tensor k, at::Tensor wq, at::Tensor wk,
                        k.stride(0), k.stride(2), k.stride(1), rows, (float)eps);
 }
 
+
+// ---------------------------------------------------------------------------
+// Attention v4: swapped-QK^T 32x32 structure (guide Appendix B ladder).
+//
+// mfma_f32_32x32x16_bf16 computing S^T = K·Q^T, so each LANE owns a full
+// query ROW: softmax max/sum are a 32-value local reduce + ONE
+// __shfl_xor(32) with the partner lane; running (m, l) and the per-tile
+// O-rescale are lane-local scalars. P converts f32->bf16 in-register
+// (v_cvt_pk_bf16_f32) and redistributes across half-waves with
+// v_permlane32_swap (T12) — NO P staging through LDS. PV computes
+// O^T = V^T · P^T (same P fragments serve as the B operand), keeping the
+// output row lane-local for the rescale and the 1/l epilogue.
+//
+// Geometry: 256-thread blocks (4 waves x 32 q-rows = 128 rows/block);
+// LDS = K[64][D+8] + swizzled V^T[D][72] = 35.8 KB -> two blocks co-resident
+// per CU, NOT barrier-synced against each other (phase overlap for free).
+// ---------------------------------------------------------------------------
+typedef __attribute__((ext_vector_type(16))) float f32x16;
+
+PA_DEV f32x16 mfma32x32x16(bf16x8 a, bf16x8 b, f32x16 c) {
+    return __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, c, 0, 0, 0);
+}
+
+PA_DEV unsigned int cvt_pk_bf16(float lo, float hi) {
+    unsigned int r;
+    asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(r) : "v"(lo), "v"(hi));
+    return r;
+}
+
+template <int D>
+__global__ __launch_bounds__(256, 2) void attn_fwd_v4_kernel(
+    const bf16* __restrict__ q, const bf16* __restrict__ k,
+    const bf16* __restrict__ v, bf16* __restrict__ out,
+    int S, int Sk, float scale, int H,
+    long q_bs, long q_hs, int q_ss,
+    long k_bs, long k_hs, int k_ss,
+    long v_bs, long v_hs, int v_ss,
+    long o_bs, long o_hs, int o_ss) {
+    constexpr int KVBLK = 64;
+    constexpr int WAVES = 4;
+    constexpr int THREADS = WAVES * 64;
+    constexpr int KPAD = D + 8;
+    constexpr int VPAD = KVBLK + 8;
+    constexpr int KK = D / 16;        // QK^T K-steps per 32-key tile
+    constexpr int NV = D / 32;        // PV dim tiles
+    constexpr int KVECS = (KVBLK * D) / (8 * THREADS);
+
+    __shared__ bf16 k_lds[KVBLK * KPAD];
+    __shared__ bf16 v_lds[D * VPAD];
+    __shared__ float l_bcast[WAVES];  // unused slot keeps LDS struct simple
+
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wid = tid >> 6;
+    const int l32 = lane & 31;        // this lane's query row (within wave)
+    const int hi = lane >> 5;         // half-wave id
+
+    const long bh = blockIdx.y;
+    const long b = bh / H;
+    const int h = (int)(bh % H);
+    const int q0 = blockIdx.x * (WAVES * 32) + wid * 32;
+
+    const bf16* qp = q + b * q_bs + (long)h * q_hs;
+    const bf16* kp = k + b * k_bs + (long)h * k_hs;
+    const bf16* vp = v + b * v_bs + (long)h * v_hs;
+    bf16* op = out + b * o_bs + (long)h * o_hs;
+
+    // Q fragments (B-operand of the swapped QK^T): lane holds
+    // Q[q0 + l32][kk*16 + hi*8 + j], j = 0..7.
+    bf16x8 qfrag[KK];
+    {
+        const int row = q0 + l32;
+        const int rr = row < S ? row : (S > 0 ? S - 1 : 0);
+#pragma unroll
+        for (int kk = 0; kk < KK; ++kk)
+            qfrag[kk] = *reinterpret_cast<const bf16x8*>(
+                qp + (long)rr * q_ss + kk * 16 + hi * 8);
+    }
+
+    f32x16 o_acc[NV];
+#pragma unroll
+    for (int n = 0; n < NV; ++n)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) o_acc[n][r] = 0.f;
+    float m_run = -1e30f, l_run = 0.f;
+    const float scale2 = scale * PA_LOG2E;
+
+    bf16x8 kreg[KVECS], vreg[KVECS];
+    auto issue_tile_loads = [&](int kv0) {
+#pragma unroll
+        for (int i = 0; i < KVECS; ++i) {
+            const int idx = tid + i * THREADS;
+            const int row = idx / (D / 8);
+            const int col = (idx % (D / 8)) * 8;
+            const int src = kv0 + row;
+            if (src < Sk) {
+                kreg[i] = *reinterpret_cast<const bf16x8*>(kp + (long)src * k_ss + col);
+                vreg[i] = *reinterpret_cast<const bf16x8*>(vp + (long)src * v_ss + col);
+            } else {
+                kreg[i] = bf16x8{0,0,0,0,0,0,0,0};
+                vreg[i] = bf16x8{0,0,0,0,0,0,0,0};
+            }
+        }
+    };
+    auto write_tile_lds = [&]() {
+#pragma unroll
+        for (int i = 0; i < KVECS; ++i) {
+            const int idx = tid + i * THREADS;
+            const int row = idx / (D / 8);
+            const int col = (idx % (D / 8)) * 8;
+            *reinterpret_cast<bf16x8*>(&k_lds[row * KPAD + col]) = kreg[i];
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                const int dim = col + j;
+                const int key_swz =
+                    (((row >> 3) ^ ((dim >> 3) & 7)) << 3) | (row & 7);
+                v_lds[dim * VPAD + key_swz] =
+                    __ushort_as_bfloat16((unsigned short)vreg[i][j]);
+            }
+        }
+    };
+
+    issue_tile_loads(0);
+    const int n_tiles = (Sk + KVBLK - 1) / KVBLK;
+    for (int t = 0; t < n_tiles; ++t) {
+        const int kv0 = t * KVBLK;
+        __syncthreads();
+        write_tile_lds();
+        __syncthreads();
+        if (t + 1 < n_tiles) issue_tile_loads(kv0 + KVBLK);
+
+        // ---- swapped QK^T: S^T[key][row] for two 32-key tiles ------------
+        f32x16 st[2];
+#pragma unroll
+        for (int kt = 0; kt < 2; ++kt) {
+#pragma unroll
+            for (int r = 0; r < 16; ++r) st[kt][r] = 0.f;
+            __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+            for (int kk = 0; kk < KK; ++kk) {
+                // A = K chunk: lane holds K[kt*32 + l32][kk*16 + hi*8 + j]
+                bf16x8 afrag = *reinterpret_cast<const bf16x8*>(
+                    &k_lds[(kt * 32 + l32) * KPAD + kk * 16 + hi * 8]);
+                st[kt] = mfma32x32x16(afrag, qfrag[kk], st[kt]);
+            }
+            __builtin_amdgcn_s_setprio(0);
+        }
+
+        // ---- lane-local online softmax (this lane's row = q0 + l32) ------
+        // value (kt, reg) = S[row][key = kt*32 + (reg&3) + 8*(reg>>2) + 4*hi]
+        float mx = -1e30f;
+#pragma unroll
+        for (int kt = 0; kt < 2; ++kt)
+#pragma unroll
+            for (int r = 0; r < 16; ++r) {
+                const int key = kv0 + kt * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+                const float sv = (key < Sk) ? st[kt][r] * scale2 : -1e30f;
+                st[kt][r] = sv;
+                mx = fmaxf(mx, sv);
+            }
+        mx = fmaxf(mx, __shfl_xor(mx, 32, 64));  // partner holds the row's other 32 keys
+        const float mnew = fmaxf(m_run, mx);
+        const float alpha = exp2f(m_run - mnew);
+        m_run = mnew;
+        float ps = 0.f;
+#pragma unroll
+        for (int kt = 0; kt < 2; ++kt)
+#pragma unroll
+            for (int r = 0; r < 16; ++r) {
+                const float pv_ = exp2f(st[kt][r] - mnew);
+                st[kt][r] = pv_;
+                ps += pv_;
+            }
+        ps += __shfl_xor(ps, 32, 64);
+        l_run = l_run * alpha + ps;
+        if (alpha != 1.f) {
+#pragma unroll
+            for (int n = 0; n < NV; ++n)
+#pragma unroll
+                for (int r = 0; r < 16; ++r) o_acc[n][r] *= alpha;
+        }
+
+        // ---- P f32 -> bf16 fragments via cvt_pk + permlane32_swap --------
+        // chunk c (16 keys) uses regs 8*(c&1)..8*(c&1)+7 of st[c>>1]; after
+        // the half-swap each lane holds P[row l32][chunk base + hi*8 + j].
+        bf16x8 pfrag[4];
+#pragma unroll
+        for (int c = 0; c < 4; ++c) {
+            const f32x16& sv = st[c >> 1];
+            const int rb = 8 * (c & 1);
+            unsigned int w0 = cvt_pk_bf16(sv[rb + 0], sv[rb + 1]);
+            unsigned int w1 = cvt_pk_bf16(sv[rb + 2], sv[rb + 3]);
+            unsigned int w2 = cvt_pk_bf16(sv[rb + 4], sv[rb + 5]);
+            unsigned int w3 = cvt_pk_bf16(sv[rb + 6], sv[rb + 7]);
+            auto r02 = __builtin_amdgcn_permlane32_swap(w0, w2, false, false);
+            auto r13 = __builtin_amdgcn_permlane32_swap(w1, w3, false, false);
+            unsigned int d[4] = {(unsigned int)r02[0], (unsigned int)r13[0],
+                                 (unsigned int)r02[1], (unsigned int)r13[1]};
+            pfrag[c] = *reinterpret_cast<bf16x8*>(d);
+        }
+
+        // ---- PV: O^T[dim][row] += V^T chunk · P^T chunk ------------------
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int c = 0; c < 4; ++c) {
+#pragma unroll
+            for (int n = 0; n < NV; ++n) {
+                // A = V^T: lane holds V^T[n*32 + l32][16c + hi*8 + j]
+                const int dim = n * 32 + l32;
+                const int gsw = ((2 * c + hi) ^ ((dim >> 3) & 7)) << 3;
+                bf16x8 va = *reinterpret_cast<const bf16x8*>(
+                    &v_lds[dim * VPAD + gsw]);
+                o_acc[n] = mfma32x32x16(va, pfrag[c], o_acc[n]);
+            }
+        }
+        __builtin_amdgcn_s_setprio(0);
+    }
+
+    // ---- epilogue: O = O^T / l, row is lane-local -------------------------
+    (void)l_bcast;
+    const int row = q0 + l32;
+    if (row < S) {
+        const float inv_l = (l_run > 0.f) ? 1.f / l_run : 0.f;
+#pragma unroll
+        for (int n = 0; n < NV; ++n) {
+#pragma unroll
+            for (int r2 = 0; r2 < 4; ++r2) {
+                // dims n*32 + 8*r2 + 4*hi + (0..3) are consecutive
+                const int dim0 = n * 32 + 8 * r2 + 4 * hi;
+                unsigned short pack[4];
+#pragma unroll
+                for (int j = 0; j < 4; ++j)
+                    pack[j] = __bfloat16_as_ushort(
+                        f2bf(o_acc[n][r2 * 4 + j] * inv_l));
+                *reinterpret_cast<unsigned long long*>(
+                    op + (long)row * o_ss + dim0) =
+                    *reinterpret_cast<unsigned long long*>(pack);
+            }
+        }
+    }
+}
+
 struct AttnStrides {
     long bs, hs;
     int ss;
@@ -958,6 +1200,11 @@ static at::Tensor attn_fwd_launch(at::Tensor q, at::Tensor k, at::Tensor v,
     at::Tensor out = bshd
         ? at::empty({B, S, H, D}, qc.options())
         : at::empty({B, H, S, D}, qc.options());
+    static const bool use_v3 = []() {
+        const char* e = getenv("PA_ATTN_V3");
+        return e && e[0] == '1';
+    }();
+    dim3 grid_v4((unsigned)((S + 127) / 128), (unsigned)((long)B * H));
     dim3 grid((unsigned)((S + 255) / 256), (unsigned)((long)B * H));
 #define PA_ATTN_ARGS                                                          \
     (const bf16*)qc.data_ptr(), (const bf16*)kc.data_ptr(),                   \
@@ -966,7 +1213,15 @@ static at::Tensor attn_fwd_launch(at::Tensor q, at::Tensor k, at::Tensor v,
     kc.stride(b_ax), kc.stride(h_ax), (int)kc.stride(s_ax),                   \
     vc.stride(b_ax), vc.stride(h_ax), (int)vc.stride(s_ax),                   \
     out.stride(b_ax), out.stride(h_ax), (int)out.stride(s_ax)
-    if (D == 128) {
+    if (!use_v3) {
+        if (D == 128) {
+            hipLaunchKernelGGL(attn_fwd_v4_kernel<128>, grid_v4, dim3(256), 0,
+                               cur_stream(), PA_ATTN_ARGS);
+        } else {
+            hipLaunchKernelGGL(attn_fwd_v4_kernel<64>, grid_v4, dim3(256), 0,
+                               cur_stream(), PA_ATTN_ARGS);
+        }
+    } else if (D == 128) {
         hipLaunchKernelGGL(attn_fwd_kernel<128>, grid, dim3(512), 0,
                            cur_stream(), PA_ATTN_ARGS);
     } else {
