@@ -55,3 +55,48 @@ def auto_split_batch(
     if not any(str(d).startswith("cuda") for d in devices):
         return compute_split_sizes(batch_size, weights)
     return compute_split_sizes(batch_size, vram_blended_weights(devices, weights))
+
+
+class AdaptiveBalancer:
+    """Per-step timing feedback on top of the static weights.
+
+    The reference's own limitation list calls out "Static load balancing -
+    Percentages fixed per run" (reference README.md); this closes it: after
+    each DP step, per-device busy times update an EMA of throughput
+    (samples/sec), and the effective weights blend user weights with the
+    measured throughput share. A device that keeps finishing late sheds load.
+    """
+
+    def __init__(self, devices: Sequence[str], user_weights: Sequence[float],
+                 blend: float = 0.5, ema: float = 0.5):
+        self.devices = list(devices)
+        self.user_weights = list(user_weights)
+        self.blend = blend
+        self.ema = ema
+        self.throughput = {d: None for d in self.devices}
+
+    def record(self, device: str, chunk_size: int, seconds: float) -> None:
+        if seconds <= 0 or chunk_size <= 0:
+            return
+        tput = chunk_size / seconds
+        prev = self.throughput.get(device)
+        self.throughput[device] = (
+            tput if prev is None else self.ema * tput + (1 - self.ema) * prev
+        )
+
+    def weights(self) -> List[float]:
+        tputs = [self.throughput.get(d) for d in self.devices]
+        if any(t is None for t in tputs):
+            return list(self.user_weights)
+        total_t = sum(tputs)
+        if total_t <= 0:
+            return list(self.user_weights)
+        blended = [
+            (1 - self.blend) * w + self.blend * (t / total_t)
+            for w, t in zip(self.user_weights, tputs)
+        ]
+        total = sum(blended)
+        return [b / total for b in blended]
+
+    def split(self, batch_size: int) -> List[int]:
+        return compute_split_sizes(batch_size, self.weights())

@@ -35,7 +35,7 @@ from typing import Any, Dict, List, Optional
 import torch
 from torch import nn
 
-from .balance import auto_split_batch
+from .balance import AdaptiveBalancer, auto_split_batch
 from .chain import DeviceChain
 from .replicate import replicate_module
 from .split import (
@@ -83,6 +83,10 @@ class ParallelEngine:
         self.streams: Dict[str, Optional[torch.cuda.Stream]] = {}
         self.pipeline = None  # set by pipeline.configure_pipeline
         self._lead_t = torch.device(chain.lead)
+        # per-step timing feedback (closes the reference's static-balancing
+        # limitation); engaged only when auto_vram_balance is on
+        self.balancer: Optional[AdaptiveBalancer] = None
+        self._pending_times: list = []
 
     # ------------------------------------------------------------------
     # Setup: replicate with OOM degradation.
@@ -154,7 +158,12 @@ class ParallelEngine:
         if not self.workload_split or batch < len(devices) or len(devices) == 1:
             return self._lead_only(x, timesteps, context, **kwargs)
 
-        if self.auto_vram_balance:
+        self._harvest_times()
+        if self.auto_vram_balance and self.balancer is not None and all(
+            self.balancer.throughput.get(d) is not None for d in devices
+        ):
+            sizes = self.balancer.split(batch)
+        elif self.auto_vram_balance:
             sizes = auto_split_batch(batch, devices, weights)
         else:
             sizes = compute_split_sizes(batch, weights)
@@ -195,7 +204,29 @@ class ParallelEngine:
                 devices, sizes, batch, x, timesteps, context, kwargs
             )
 
+    def _harvest_times(self):
+        if not self._pending_times:
+            return
+        done = []
+        for dev, size, rec in self._pending_times:
+            if isinstance(rec, float):
+                self.balancer.record(dev, size, rec)
+                done.append((dev, size, rec))
+            else:
+                ev0, ev1 = rec
+                try:
+                    if ev1.query():
+                        self.balancer.record(
+                            dev, size, ev0.elapsed_time(ev1) / 1000.0
+                        )
+                        done.append((dev, size, rec))
+                except Exception:  # noqa: BLE001
+                    done.append((dev, size, rec))
+        self._pending_times = [p for p in self._pending_times if p not in done]
+
     def _data_parallel_impl(self, devices, sizes, batch, x, timesteps, context, kwargs):
+        if self.balancer is None and self.auto_vram_balance:
+            self.balancer = AdaptiveBalancer(devices, list(self.chain.weights))
         x_chunks = split_batch(x, sizes)
         t_chunks = split_batch(timesteps, sizes)
         c_chunks = split_batch(context, sizes) if context is not None else None
@@ -218,6 +249,9 @@ class ParallelEngine:
             stream = self.streams.get(dev)
             try:
                 if stream is None:  # cpu worker: synchronous
+                    import time as _time
+
+                    _t0 = _time.perf_counter()
                     out = self._run_chunk(
                         dev,
                         move_to_device(x_chunks[i], dev),
@@ -226,6 +260,10 @@ class ParallelEngine:
                         {k: move_to_device(v, dev)
                          for k, v in kw_chunks[i].items()},
                     )
+                    if self.balancer is not None:
+                        self._pending_times.append(
+                            (dev, sizes[i], _time.perf_counter() - _t0)
+                        )
                     results[i] = move_to_device(out, lead)
                     continue
                 with torch.cuda.stream(stream), self._lead_stream_ctx(lead_stream):
@@ -248,7 +286,15 @@ class ParallelEngine:
                         ev_in = torch.cuda.Event()
                         ev_in.record(lead_stream)
                         stream.wait_event(ev_in)
+                    ev0 = None
+                    if self.balancer is not None:
+                        ev0 = torch.cuda.Event(enable_timing=True)
+                        ev0.record(stream)
                     out = self._run_chunk(dev, xi, ti, ci, kwi)
+                    if ev0 is not None:
+                        ev1 = torch.cuda.Event(enable_timing=True)
+                        ev1.record(stream)
+                        self._pending_times.append((dev, sizes[i], (ev0, ev1)))
                     out_lead = move_to_device(out, lead, non_blocking=True)
                     ev = torch.cuda.Event()
                     ev.record(stream)

@@ -70,3 +70,40 @@ def test_gpu_policy_keeps_fp8():
     t = torch.zeros(4, dtype=torch.float8_e4m3fn)
     out = fp8.sanitize_param_dtype(t, "cuda:0")
     assert out.dtype == torch.float8_e4m3fn
+
+
+def test_adaptive_balancer_shifts_load():
+    from comfyui_parallelanything_amd.parallel.balance import AdaptiveBalancer
+
+    b = AdaptiveBalancer(["a", "b"], [0.5, 0.5], blend=0.5, ema=1.0)
+    assert b.weights() == [0.5, 0.5]  # no data yet -> user weights
+    # device a is 3x faster
+    for _ in range(3):
+        b.record("a", 6, 1.0)
+        b.record("b", 2, 1.0)
+    ws = b.weights()
+    # 0.5*0.5 + 0.5*0.75 = 0.625 vs 0.375
+    assert ws[0] == pytest.approx(0.625)
+    sizes = b.split(8)
+    assert sizes[0] == 5 and sizes[1] == 3
+
+
+def test_adaptive_balancer_in_engine():
+    import torch
+
+    from comfyui_parallelanything_amd.models.registry import make_sd15, sd15_inputs
+    from comfyui_parallelanything_amd.parallel.chain import DeviceChain, make_entry
+    from comfyui_parallelanything_amd.parallel.engine import ParallelEngine
+
+    m = make_sd15(tiny=True)
+    eng = ParallelEngine(
+        DeviceChain.from_list([make_entry("cpu", 50), make_entry("cpu", 50)]),
+        auto_vram_balance=True,
+    )
+    eng.setup(m)
+    x, t, c, kw = sd15_inputs(4, tiny=True)
+    ref = m(x, t, context=c, **kw)
+    for _ in range(3):
+        out = eng.forward(x, t, context=c, **kw)
+    torch.testing.assert_close(out, ref, rtol=1e-4, atol=1e-5)
+    assert eng.balancer is not None
