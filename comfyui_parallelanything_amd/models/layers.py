@@ -22,6 +22,13 @@ from torch import nn
 from .. import ops
 
 
+class FusedGELU(nn.Module):
+    """tanh-GELU through the ops dispatch (vectorized gfx950 kernel)."""
+
+    def forward(self, x):
+        return ops.gelu_tanh(x)
+
+
 class MLPEmbedder(nn.Module):
     """2-layer SiLU MLP used for timestep / vector conditioning."""
 
@@ -144,7 +151,7 @@ class DoubleStreamBlock(nn.Module):
         self.img_attn_norm = QKNorm(head_dim)
         self.img_attn_proj = nn.Linear(hidden, hidden)
         self.img_mlp = nn.Sequential(
-            nn.Linear(hidden, mlp_dim), nn.GELU(approximate="tanh"),
+            nn.Linear(hidden, mlp_dim), FusedGELU(),
             nn.Linear(mlp_dim, hidden),
         )
         self.txt_mod = Modulation(hidden, double=True)
@@ -152,7 +159,7 @@ class DoubleStreamBlock(nn.Module):
         self.txt_attn_norm = QKNorm(head_dim)
         self.txt_attn_proj = nn.Linear(hidden, hidden)
         self.txt_mlp = nn.Sequential(
-            nn.Linear(hidden, mlp_dim), nn.GELU(approximate="tanh"),
+            nn.Linear(hidden, mlp_dim), FusedGELU(),
             nn.Linear(mlp_dim, hidden),
         )
         self.attn = JointAttention(num_heads, head_dim)
@@ -216,7 +223,7 @@ class SingleStreamBlock(nn.Module):
         self.norm = QKNorm(head_dim)
         self.modulation = Modulation(hidden, double=False)
         self.attn = JointAttention(num_heads, head_dim)
-        self.mlp_act = nn.GELU(approximate="tanh")
+        self.mlp_act = FusedGELU()
 
     def forward(self, x, vec, pe):
         B, S, hidden = x.shape

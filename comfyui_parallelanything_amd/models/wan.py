@@ -16,7 +16,7 @@ import torch
 from torch import nn
 
 from .. import ops
-from .layers import MLPEmbedder, QKNorm
+from .layers import FusedGELU, MLPEmbedder, QKNorm
 
 
 def rope_3d_table(f: int, h: int, w: int, axes_dim: Tuple[int, ...],
@@ -48,7 +48,7 @@ class WanBlock(nn.Module):
         self.cross_v = nn.Linear(ctx_dim, dim)
         self.cross_proj = nn.Linear(dim, dim)
         self.ffn = nn.Sequential(
-            nn.Linear(dim, ffn_dim), nn.GELU(approximate="tanh"),
+            nn.Linear(dim, ffn_dim), FusedGELU(),
             nn.Linear(ffn_dim, dim),
         )
 
@@ -119,7 +119,7 @@ class WanDiT(nn.Module):
         self.patch_dim = cfg.in_channels * pf * ph * pw
         self.patch_in = nn.Linear(self.patch_dim, cfg.dim)
         self.txt_in = nn.Sequential(
-            nn.Linear(cfg.ctx_dim, cfg.dim), nn.GELU(approximate="tanh"),
+            nn.Linear(cfg.ctx_dim, cfg.dim), FusedGELU(),
             nn.Linear(cfg.dim, cfg.dim),
         )
         self.time_in = MLPEmbedder(cfg.time_embed_dim, cfg.dim)

@@ -228,6 +228,14 @@ def qk_norm_rope_(q, k, wq, wk, cs, eps: float = 1e-6):
     return q, k
 
 
+def gelu_tanh(x: torch.Tensor) -> torch.Tensor:
+    if x.is_cuda:
+        ext = _require_ext("gelu_tanh")
+        if ext is not None:
+            return ext.gelu_tanh(x)
+    return torch.nn.functional.gelu(x, approximate="tanh")
+
+
 def timestep_embedding(t, dim: int, max_period: float = 10000.0,
                        time_factor: float = 1000.0) -> torch.Tensor:
     if t.is_cuda:

@@ -290,6 +290,28 @@ __global__ void gate_residual_bf16_kernel(const bf16* __restrict__ res,
     }
 }
 
+
+// Vectorized tanh-GELU (bf16 short8): the MLP activation between the two
+// hipBLASLt GEMMs ([B, S, 4*hidden] tensors; memory-bound).
+__global__ void gelu_tanh_bf16_kernel(const bf16* __restrict__ x,
+                                      bf16* __restrict__ out, long total8) {
+    const long stride = (long)gridDim.x * blockDim.x;
+    const short8* xv = reinterpret_cast<const short8*>(x);
+    short8* ov = reinterpret_cast<short8*>(out);
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total8;
+         i += stride) {
+        short8 v = xv[i], o;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            const float f = bf2f(__ushort_as_bfloat16((unsigned short)v[j]));
+            const float c = 0.7978845608028654f * (f + 0.044715f * f * f * f);
+            const float g = 0.5f * f * (1.f + tanhf(c));
+            o[j] = (short)__bfloat16_as_ushort(f2bf(g));
+        }
+        ov[i] = o;
+    }
+}
+
 // ---------------------------------------------------------------------------
 // GroupNorm + SiLU: x [B, C, H, W]; one block per (b, group).
 // ---------------------------------------------------------------------------
@@ -1164,6 +1186,22 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_v4_kernel(
     }
 }
 
+
+at::Tensor gelu_tanh(at::Tensor x) {
+    CHECK_GPU(x);
+    auto xc = x.contiguous();
+    auto out = at::empty_like(xc);
+    if (xc.scalar_type() == at::kBFloat16 && (xc.numel() % 8) == 0) {
+        const long total8 = xc.numel() / 8;
+        const int blocks = (int)std::min<long>((total8 + 255) / 256, 4096);
+        hipLaunchKernelGGL(gelu_tanh_bf16_kernel, dim3(blocks), dim3(256), 0,
+                           cur_stream(), (const bf16*)xc.data_ptr(),
+                           (bf16*)out.data_ptr(), total8);
+        return out;
+    }
+    return at::gelu(xc, "tanh");
+}
+
 struct AttnStrides {
     long bs, hs;
     int ss;
@@ -1253,6 +1291,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("attn_fwd", &attn_fwd, "Fused flash attention fwd, bf16 MFMA (gfx950)");
     m.def("attn_fwd_bshd", &attn_fwd_bshd,
           "Fused flash attention fwd on [B,S,H,D] strided views (gfx950)");
+    m.def("gelu_tanh", &gelu_tanh, "Vectorized tanh-GELU (gfx950)");
     m.def("qk_norm_rope_", &qk_norm_rope_,
           "In-place fused qk RMSNorm + RoPE on [B,S,H,D] views (gfx950)");
 }
