@@ -182,15 +182,19 @@ class DoubleStreamBlock(nn.Module):
         img_in = ops.layer_norm_mod(img, img_m1.scale, img_m1.shift)
         txt_in = ops.layer_norm_mod(txt, txt_m1.scale, txt_m1.shift)
         T = txt.shape[1]
-        iq, ik, iv = self._qkv(img_in, self.img_attn_qkv, self.img_attn_norm,
-                               pe[T:])
-        tq, tk, tv = self._qkv(txt_in, self.txt_attn_qkv, self.txt_attn_norm,
-                               pe[:T])
-
-        # joint sequence: txt first, then img (FLUX convention)
-        q = torch.cat([tq, iq], dim=1)
-        k = torch.cat([tk, ik], dim=1)
-        v = torch.cat([tv, iv], dim=1)
+        H = self.num_heads
+        txt_qkv = self.txt_attn_qkv(txt_in).unflatten(-1, (3, H, -1))
+        img_qkv = self.img_attn_qkv(img_in).unflatten(-1, (3, H, -1))
+        # fused: per-stream qk-norm + RoPE + contiguous joint q/k/v
+        # (txt first, then img — FLUX convention)
+        q, k, v = ops.pack_joint_qkv(
+            txt_qkv, img_qkv,
+            self.txt_attn_norm.query_norm.scale,
+            self.txt_attn_norm.key_norm.scale,
+            self.img_attn_norm.query_norm.scale,
+            self.img_attn_norm.key_norm.scale,
+            pe,
+        )
         attn = ops.attention_bshd(q, k, v, self.attn.scale).flatten(2)
         txt_attn, img_attn = attn[:, :T], attn[:, T:]
 

@@ -228,6 +228,28 @@ def qk_norm_rope_(q, k, wq, wk, cs, eps: float = 1e-6):
     return q, k
 
 
+def pack_joint_qkv(txt_qkv, img_qkv, wq_t, wk_t, wq_i, wk_i, cs,
+                   eps: float = 1e-6):
+    """Dual-stream joint qkv pack: per-stream qk RMSNorm + RoPE + contiguous
+    [B, T+Si, H, D] q/k/v in one pass (txt_qkv/img_qkv: [B,S,3,H,D] views)."""
+    if txt_qkv.is_cuda:
+        ext = _require_ext("pack_joint_qkv")
+        if ext is not None and txt_qkv.dtype == torch.bfloat16                 and txt_qkv.shape[-1] <= 128:
+            return ext.pack_joint_qkv(txt_qkv, img_qkv, wq_t, wk_t, wq_i, wk_i,
+                                      cs, eps)
+        _unsupported("pack_joint_qkv",
+                     f"dtype={txt_qkv.dtype}, D={txt_qkv.shape[-1]}")
+    T = txt_qkv.shape[1]
+    tq, tk, tv = txt_qkv.unbind(2)
+    iq, ik, iv = img_qkv.unbind(2)
+    qk_norm_rope_(tq, tk, wq_t, wk_t, cs[:T], eps)
+    qk_norm_rope_(iq, ik, wq_i, wk_i, cs[T:], eps)
+    q = torch.cat([tq, iq], dim=1)
+    k = torch.cat([tk, ik], dim=1)
+    v = torch.cat([tv, iv], dim=1)
+    return q, k, v
+
+
 def gelu_tanh(x: torch.Tensor) -> torch.Tensor:
     if x.is_cuda:
         ext = _require_ext("gelu_tanh")

@@ -14,6 +14,7 @@
 // Build: ops/build.py (hipcc --offload-arch=gfx950, in-tree .so).
 
 #include <torch/extension.h>
+#include <vector>
 #include <ATen/ATen.h>
 #include <c10/hip/HIPStream.h>
 
@@ -945,6 +946,81 @@ void qk_norm_rope_(at::Tensor q, at::Tensor k, at::Tensor wq, at::Tensor wk,
 }
 
 
+
+// ---------------------------------------------------------------------------
+// Fused joint-qkv pack for dual-stream MMDiT blocks: reads the txt and img
+// qkv projection buffers (strided [B,S,3,H,D] views), applies per-stream qk
+// RMSNorm + RoPE, and writes CONTIGUOUS joint q/k/v [B, T+Si, H, D] —
+// replacing three strided torch.cat copies plus two qk_norm_rope passes
+// with a single bandwidth pass. One wave per (b, s, h); lane = pair index.
+// ---------------------------------------------------------------------------
+__global__ void pack_joint_qkv_kernel(
+    const bf16* __restrict__ txt, const bf16* __restrict__ img,
+    const bf16* __restrict__ wq_t, const bf16* __restrict__ wk_t,
+    const bf16* __restrict__ wq_i, const bf16* __restrict__ wk_i,
+    const float* __restrict__ cs,                   // [T+Si][D/2][2]
+    bf16* __restrict__ oq, bf16* __restrict__ ok, bf16* __restrict__ ov,
+    int T, int Si, int H, int D,
+    long t_bs, long t_ss, long t_qs, long t_hs,     // txt strides (b, s, qkv, h)
+    long i_bs, long i_ss, long i_qs, long i_hs,
+    long n_rows, float eps) {
+    const int lane = threadIdx.x & 63;
+    const long row = ((long)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+    if (row >= n_rows) return;
+    const int S = T + Si;
+    const long b = row / ((long)S * H);
+    const long sh = row % ((long)S * H);
+    const int sj = (int)(sh / H);
+    const int h = (int)(sh % H);
+    const int pairs = D / 2;
+    if (lane >= pairs) return;
+
+    const bool is_txt = sj < T;
+    const bf16* src = is_txt
+        ? txt + b * t_bs + (long)sj * t_ss + (long)h * t_hs
+        : img + b * i_bs + (long)(sj - T) * i_ss + (long)h * i_hs;
+    const long qs = is_txt ? t_qs : i_qs;
+    const unsigned int uwq = reinterpret_cast<const unsigned int*>(
+        is_txt ? wq_t : wq_i)[lane];
+    const unsigned int uwk = reinterpret_cast<const unsigned int*>(
+        is_txt ? wk_t : wk_i)[lane];
+    const float c = cs[((long)sj * pairs + lane) * 2 + 0];
+    const float sn = cs[((long)sj * pairs + lane) * 2 + 1];
+
+    const unsigned int uq = reinterpret_cast<const unsigned int*>(src)[lane];
+    const unsigned int uk = reinterpret_cast<const unsigned int*>(src + qs)[lane];
+    const unsigned int uv = reinterpret_cast<const unsigned int*>(src + 2 * qs)[lane];
+
+    const long obase = (((long)b * S + sj) * H + h) * (D / 2);
+    {   // q: rms + weight + rope
+        float a0 = bf2f(__ushort_as_bfloat16((unsigned short)(uq & 0xffff)));
+        float a1 = bf2f(__ushort_as_bfloat16((unsigned short)(uq >> 16)));
+        float ss_ = a0 * a0 + a1 * a1;
+#pragma unroll
+        for (int off = 32; off > 0; off >>= 1) ss_ += __shfl_xor(ss_, off, 64);
+        const float rr = rsqrtf(ss_ / (float)D + eps);
+        a0 = a0 * rr * bf2f(__ushort_as_bfloat16((unsigned short)(uwq & 0xffff)));
+        a1 = a1 * rr * bf2f(__ushort_as_bfloat16((unsigned short)(uwq >> 16)));
+        reinterpret_cast<unsigned int*>(oq)[obase + lane] =
+            (unsigned int)__bfloat16_as_ushort(f2bf(a0 * c - a1 * sn)) |
+            ((unsigned int)__bfloat16_as_ushort(f2bf(a0 * sn + a1 * c)) << 16);
+    }
+    {   // k
+        float a0 = bf2f(__ushort_as_bfloat16((unsigned short)(uk & 0xffff)));
+        float a1 = bf2f(__ushort_as_bfloat16((unsigned short)(uk >> 16)));
+        float ss_ = a0 * a0 + a1 * a1;
+#pragma unroll
+        for (int off = 32; off > 0; off >>= 1) ss_ += __shfl_xor(ss_, off, 64);
+        const float rr = rsqrtf(ss_ / (float)D + eps);
+        a0 = a0 * rr * bf2f(__ushort_as_bfloat16((unsigned short)(uwk & 0xffff)));
+        a1 = a1 * rr * bf2f(__ushort_as_bfloat16((unsigned short)(uwk >> 16)));
+        reinterpret_cast<unsigned int*>(ok)[obase + lane] =
+            (unsigned int)__bfloat16_as_ushort(f2bf(a0 * c - a1 * sn)) |
+            ((unsigned int)__bfloat16_as_ushort(f2bf(a0 * sn + a1 * c)) << 16);
+    }
+    reinterpret_cast<unsigned int*>(ov)[obase + lane] = uv;  // v passthrough
+}
+
 // ---------------------------------------------------------------------------
 // Attention v4: swapped-QK^T 32x32 structure (guide Appendix B ladder).
 //
@@ -1202,6 +1278,48 @@ at::Tensor gelu_tanh(at::Tensor x) {
     return at::gelu(xc, "tanh");
 }
 
+
+std::vector<at::Tensor> pack_joint_qkv(at::Tensor txt_qkv, at::Tensor img_qkv,
+                                       at::Tensor wq_t, at::Tensor wk_t,
+                                       at::Tensor wq_i, at::Tensor wk_i,
+                                       at::Tensor cs, double eps) {
+    CHECK_GPU(txt_qkv);
+    TORCH_CHECK(txt_qkv.dim() == 5 && img_qkv.dim() == 5,
+                "pack_joint_qkv expects [B,S,3,H,D] views");
+    TORCH_CHECK(txt_qkv.scalar_type() == at::kBFloat16, "bf16 only");
+    const int B = (int)txt_qkv.size(0), T = (int)txt_qkv.size(1),
+              H = (int)txt_qkv.size(3), D = (int)txt_qkv.size(4);
+    const int Si = (int)img_qkv.size(1);
+    TORCH_CHECK(txt_qkv.stride(4) == 1 && img_qkv.stride(4) == 1,
+                "last dim contiguous");
+    TORCH_CHECK(D % 2 == 0 && D <= 128, "D must be even and <= 128");
+    auto csf = cs.to(at::kFloat).contiguous();
+    TORCH_CHECK((int)csf.size(0) == T + Si, "cs must cover the joint sequence");
+    auto opts = txt_qkv.options();
+    auto oq = at::empty({B, T + Si, H, D}, opts);
+    auto ok = at::empty({B, T + Si, H, D}, opts);
+    auto ov = at::empty({B, T + Si, H, D}, opts);
+    const long rows = (long)B * (T + Si) * H;
+    const long blocks = (rows * 64 + 255) / 256;
+    hipLaunchKernelGGL(pack_joint_qkv_kernel, dim3((unsigned)blocks), dim3(256),
+                       0, cur_stream(),
+                       (const bf16*)txt_qkv.data_ptr(),
+                       (const bf16*)img_qkv.data_ptr(),
+                       (const bf16*)wq_t.contiguous().data_ptr(),
+                       (const bf16*)wk_t.contiguous().data_ptr(),
+                       (const bf16*)wq_i.contiguous().data_ptr(),
+                       (const bf16*)wk_i.contiguous().data_ptr(),
+                       csf.data_ptr<float>(),
+                       (bf16*)oq.data_ptr(), (bf16*)ok.data_ptr(),
+                       (bf16*)ov.data_ptr(), T, Si, H, D,
+                       txt_qkv.stride(0), txt_qkv.stride(1), txt_qkv.stride(2),
+                       txt_qkv.stride(3),
+                       img_qkv.stride(0), img_qkv.stride(1), img_qkv.stride(2),
+                       img_qkv.stride(3),
+                       rows, (float)eps);
+    return {oq, ok, ov};
+}
+
 struct AttnStrides {
     long bs, hs;
     int ss;
@@ -1292,6 +1410,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("attn_fwd_bshd", &attn_fwd_bshd,
           "Fused flash attention fwd on [B,S,H,D] strided views (gfx950)");
     m.def("gelu_tanh", &gelu_tanh, "Vectorized tanh-GELU (gfx950)");
+    m.def("pack_joint_qkv", &pack_joint_qkv,
+          "Fused dual-stream qkv pack + qk-norm + RoPE (gfx950)");
     m.def("qk_norm_rope_", &qk_norm_rope_,
           "In-place fused qk RMSNorm + RoPE on [B,S,H,D] views (gfx950)");
 }

@@ -220,3 +220,37 @@ def test_attn_pad_head_dims():
         ).permute(0, 2, 1, 3)
         ref = R.attention(q.float(), k.float(), v.float())
         _cmp(out, ref, 2e-2, 2e-2, f"attn pad D={D}")
+
+
+def test_pack_joint_qkv_fused():
+    torch.manual_seed(7)
+    B, T, Si, H, D = 2, 16, 48, 4, 128
+    txt_qkv = torch.randn(B, T, 3, H, D, device="cuda", dtype=torch.bfloat16)
+    img_qkv = torch.randn(B, Si, 3, H, D, device="cuda", dtype=torch.bfloat16)
+    wq_t = torch.randn(D, device="cuda", dtype=torch.bfloat16)
+    wk_t = torch.randn(D, device="cuda", dtype=torch.bfloat16)
+    wq_i = torch.randn(D, device="cuda", dtype=torch.bfloat16)
+    wk_i = torch.randn(D, device="cuda", dtype=torch.bfloat16)
+    cs = R.rope_freqs(torch.arange(T + Si, device="cuda"), D)
+    q, k, v = ops.pack_joint_qkv(txt_qkv, img_qkv, wq_t, wk_t, wq_i, wk_i, cs)
+    # reference composition
+    def prep(qkv, wq, wk, cs_slice):
+        qq, kk, vv = (t.permute(0, 2, 1, 3).float() for t in qkv.unbind(2))
+        qq = R.rope_apply(R.rms_norm(qq, wq.float()), cs_slice)
+        kk = R.rope_apply(R.rms_norm(kk, wk.float()), cs_slice)
+        return qq, kk, vv
+    tq, tk, tv = prep(txt_qkv, wq_t, wk_t, cs[:T])
+    iq, ik, iv = prep(img_qkv, wq_i, wk_i, cs[T:])
+    ref_q = torch.cat([tq, iq], dim=2).permute(0, 2, 1, 3)
+    ref_k = torch.cat([tk, ik], dim=2).permute(0, 2, 1, 3)
+    ref_v = torch.cat([tv, iv], dim=2).permute(0, 2, 1, 3)
+    _cmp(q, ref_q, 3e-2, 3e-2, "pack q")
+    _cmp(k, ref_k, 3e-2, 3e-2, "pack k")
+    _cmp(v, ref_v, 0, 0, "pack v")  # bitwise passthrough
+
+
+def test_gelu_tanh_kernel():
+    x = torch.randn(3, 1000, 8, device="cuda", dtype=torch.bfloat16)
+    out = ops.gelu_tanh(x)
+    ref = torch.nn.functional.gelu(x.float(), approximate="tanh")
+    _cmp(out, ref, 2e-2, 2e-2, "gelu")
