@@ -1077,10 +1077,24 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_v4_kernel(
     const int l32 = lane & 31;        // this lane's query row (within wave)
     const int hi = lane >> 5;         // half-wave id
 
-    const long bh = blockIdx.y;
+    long bh;
+    int qtile;
+    if (gridDim.y == 1) {
+        // XCD-affine decode: xcd = id%8, bh = xcd + 8*(id/8 / nq),
+        // qtile = (id/8) % nq  (bijective when BH % 8 == 0)
+        const int nq = (S + WAVES * 32 - 1) / (WAVES * 32);
+        const long id = blockIdx.x;
+        const long xcd = id & 7;
+        const long within = id >> 3;
+        bh = xcd + 8 * (within / nq);
+        qtile = (int)(within % nq);
+    } else {
+        bh = blockIdx.y;
+        qtile = blockIdx.x;
+    }
     const long b = bh / H;
     const int h = (int)(bh % H);
-    const int q0 = blockIdx.x * (WAVES * 32) + wid * 32;
+    const int q0 = qtile * (WAVES * 32) + wid * 32;
 
     const bf16* qp = q + b * q_bs + (long)h * q_hs;
     const bf16* kp = k + b * k_bs + (long)h * k_hs;
@@ -1370,6 +1384,12 @@ static at::Tensor attn_fwd_launch(at::Tensor q, at::Tensor k, at::Tensor v,
     vc.stride(b_ax), vc.stride(h_ax), (int)vc.stride(s_ax),                   \
     out.stride(b_ax), out.stride(h_ax), (int)out.stride(s_ax)
     if (!use_v3) {
+        const long BH = (long)B * H;
+        const unsigned nq = (unsigned)((S + 127) / 128);
+        if (BH % 8 == 0) {
+            // 1-D XCD-affine grid: same-(b,h) q-tiles share an XCD's L2
+            grid_v4 = dim3((unsigned)(nq * BH), 1);
+        }
         if (D == 128) {
             hipLaunchKernelGGL(attn_fwd_v4_kernel<128>, grid_v4, dim3(256), 0,
                                cur_stream(), PA_ATTN_ARGS);
