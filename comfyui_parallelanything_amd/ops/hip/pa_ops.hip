@@ -1057,7 +1057,7 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_v4_kernel(
     long q_bs, long q_hs, int q_ss,
     long k_bs, long k_hs, int k_ss,
     long v_bs, long v_hs, int v_ss,
-    long o_bs, long o_hs, int o_ss) {
+    long o_bs, long o_hs, int o_ss, int xcd_grid) {
     constexpr int KVBLK = 64;
     constexpr int WAVES = 4;
     constexpr int THREADS = WAVES * 64;
@@ -1079,7 +1079,7 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_v4_kernel(
 
     long bh;
     int qtile;
-    if (gridDim.y == 1) {
+    if (xcd_grid) {
         // XCD-affine decode: xcd = id%8, bh = xcd + 8*(id/8 / nq),
         // qtile = (id/8) % nq  (bijective when BH % 8 == 0)
         const int nq = (S + WAVES * 32 - 1) / (WAVES * 32);
@@ -1374,6 +1374,7 @@ static at::Tensor attn_fwd_launch(at::Tensor q, at::Tensor k, at::Tensor v,
         const char* e = getenv("PA_ATTN_V3");
         return e && e[0] == '1';
     }();
+    int xcd_grid = 0;
     dim3 grid_v4((unsigned)((S + 127) / 128), (unsigned)((long)B * H));
     dim3 grid((unsigned)((S + 255) / 256), (unsigned)((long)B * H));
 #define PA_ATTN_ARGS                                                          \
@@ -1389,13 +1390,14 @@ static at::Tensor attn_fwd_launch(at::Tensor q, at::Tensor k, at::Tensor v,
         if (BH % 8 == 0) {
             // 1-D XCD-affine grid: same-(b,h) q-tiles share an XCD's L2
             grid_v4 = dim3((unsigned)(nq * BH), 1);
+            xcd_grid = 1;
         }
         if (D == 128) {
             hipLaunchKernelGGL(attn_fwd_v4_kernel<128>, grid_v4, dim3(256), 0,
-                               cur_stream(), PA_ATTN_ARGS);
+                               cur_stream(), PA_ATTN_ARGS, xcd_grid);
         } else {
             hipLaunchKernelGGL(attn_fwd_v4_kernel<64>, grid_v4, dim3(256), 0,
-                               cur_stream(), PA_ATTN_ARGS);
+                               cur_stream(), PA_ATTN_ARGS, xcd_grid);
         }
     } else if (D == 128) {
         hipLaunchKernelGGL(attn_fwd_kernel<128>, grid, dim3(512), 0,

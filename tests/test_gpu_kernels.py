@@ -254,3 +254,14 @@ def test_gelu_tanh_kernel():
     out = ops.gelu_tanh(x)
     ref = torch.nn.functional.gelu(x.float(), approximate="tanh")
     _cmp(out, ref, 2e-2, 2e-2, "gelu")
+
+
+def test_attn_single_head_batch():
+    """B*H == 1 must not take the XCD-decoded grid (regression)."""
+    torch.manual_seed(9)
+    q = torch.randn(1, 1, 300, 128, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn_like(q)
+    v = torch.randn_like(q)
+    out = ops.attention(q, k, v)
+    ref = R.attention(q.float(), k.float(), v.float())
+    _cmp(out, ref, 2e-2, 2e-2, "attn BH=1")
