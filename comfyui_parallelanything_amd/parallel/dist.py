@@ -19,11 +19,36 @@ from __future__ import annotations
 
 import datetime
 import os
-from dataclasses import dataclass
-from typing import List, Optional, Sequence
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional, Sequence
 
 import torch
 import torch.distributed as dist
+
+
+class CommStats:
+    """Per-process byte counters for the xGMI transport (SURVEY.md §5
+    metrics: 'per-GPU busy time, xGMI bytes')."""
+
+    def __init__(self):
+        self.sent_bytes = 0
+        self.recv_bytes = 0
+        self.ops = 0
+
+    def sent(self, t: torch.Tensor):
+        self.sent_bytes += t.numel() * t.element_size()
+        self.ops += 1
+
+    def recvd(self, t: torch.Tensor):
+        self.recv_bytes += t.numel() * t.element_size()
+        self.ops += 1
+
+    def summary(self) -> dict:
+        return {"sent_bytes": self.sent_bytes, "recv_bytes": self.recv_bytes,
+                "p2p_ops": self.ops}
+
+
+COMM_STATS = CommStats()
 
 
 @dataclass
@@ -99,11 +124,11 @@ def scatterv(
         assert full is not None and full.shape[0] == sum(sizes)
         offs = _offsets(sizes)
         chunks = [full[offs[i] : offs[i + 1]].contiguous() for i in range(len(sizes))]
-        ops = [
-            dist.P2POp(dist.isend, chunks[r], peer=r)
-            for r in range(info.world_size)
-            if r != src and sizes[r] > 0
-        ]
+        ops = []
+        for r in range(info.world_size):
+            if r != src and sizes[r] > 0:
+                ops.append(dist.P2POp(dist.isend, chunks[r], peer=r))
+                COMM_STATS.sent(chunks[r])
         if ops:
             for w in dist.batch_isend_irecv(ops):
                 w.wait()
@@ -114,6 +139,7 @@ def scatterv(
         (sizes[info.rank], *ref.shape[1:]), dtype=ref.dtype, device=info.device
     )
     if sizes[info.rank] > 0:
+        COMM_STATS.recvd(my)
         for w in dist.batch_isend_irecv([dist.P2POp(dist.irecv, my, peer=src)]):
             w.wait()
     return my
@@ -148,6 +174,7 @@ def gatherv(
                 buf = torch.empty_like(view)
                 views.append((view, buf))
                 ops.append(dist.P2POp(dist.irecv, buf, peer=r))
+                COMM_STATS.recvd(buf)
         if ops:
             for w in dist.batch_isend_irecv(ops):
                 w.wait()
@@ -155,9 +182,9 @@ def gatherv(
             view.copy_(buf)
         return out
     if sizes[info.rank] > 0:
-        for w in dist.batch_isend_irecv(
-            [dist.P2POp(dist.isend, chunk.contiguous(), peer=dst)]
-        ):
+        c = chunk.contiguous()
+        COMM_STATS.sent(c)
+        for w in dist.batch_isend_irecv([dist.P2POp(dist.isend, c, peer=dst)]):
             w.wait()
     return None
 

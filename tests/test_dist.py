@@ -78,3 +78,23 @@ def _all_max_worker(rank, world, tmpdir):
 )
 def test_world2_gloo(worker, tmp_path):
     mp.spawn(worker, args=(2, str(tmp_path)), nprocs=2, join=True)
+
+
+def _comm_stats_worker(rank, world, tmpdir):
+    from comfyui_parallelanything_amd.parallel.dist import (
+        COMM_STATS, gatherv, scatterv,
+    )
+
+    info = _init(rank, world, tmpdir)
+    full = torch.zeros(4, 8)
+    chunk = scatterv(full if rank == 0 else None, [2, 2], info, template=full)
+    gatherv(chunk, [2, 2], info, dst=0)
+    s = COMM_STATS.summary()
+    # each rank either sends or receives 2*8 floats per direction
+    assert s["sent_bytes"] + s["recv_bytes"] == 2 * 2 * 8 * 4
+    assert s["p2p_ops"] == 2
+    dist.destroy_process_group()
+
+
+def test_comm_stats(tmp_path):
+    mp.spawn(_comm_stats_worker, args=(2, str(tmp_path)), nprocs=2, join=True)
