@@ -15,7 +15,6 @@
 
 #include <torch/extension.h>
 #include <vector>
-#include <type_traits>
 #include <ATen/ATen.h>
 #include <c10/hip/HIPStream.h>
 
@@ -1123,34 +1122,20 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_v4_kernel(
     const float scale2 = scale * PA_LOG2E;
 
     bf16x8 kreg[KVECS], vreg[KVECS];
-    // per-thread running source pointers: advance by KVBLK rows per tile
-    // instead of re-deriving 64-bit addresses each load (saves ~150 scalar
-    // ops/tile of address arithmetic seen in the emitted .s)
-    const bf16* kptr[KVECS];
-    const bf16* vptr[KVECS];
-    int load_row[KVECS];
-#pragma unroll
-    for (int i = 0; i < KVECS; ++i) {
-        const int idx = tid + i * THREADS;
-        const int row = idx / (D / 8);
-        const int col = (idx % (D / 8)) * 8;
-        load_row[i] = row;
-        kptr[i] = kp + (long)row * k_ss + col;
-        vptr[i] = vp + (long)row * v_ss + col;
-    }
-
-    auto issue_tile_loads = [&](int kv0, bool guard) {
+    auto issue_tile_loads = [&](int kv0) {
 #pragma unroll
         for (int i = 0; i < KVECS; ++i) {
-            if (!guard || kv0 + load_row[i] < Sk) {
-                kreg[i] = *reinterpret_cast<const bf16x8*>(kptr[i]);
-                vreg[i] = *reinterpret_cast<const bf16x8*>(vptr[i]);
+            const int idx = tid + i * THREADS;
+            const int row = idx / (D / 8);
+            const int col = (idx % (D / 8)) * 8;
+            const int src = kv0 + row;
+            if (src < Sk) {
+                kreg[i] = *reinterpret_cast<const bf16x8*>(kp + (long)src * k_ss + col);
+                vreg[i] = *reinterpret_cast<const bf16x8*>(vp + (long)src * v_ss + col);
             } else {
                 kreg[i] = bf16x8{0,0,0,0,0,0,0,0};
                 vreg[i] = bf16x8{0,0,0,0,0,0,0,0};
             }
-            kptr[i] += (long)KVBLK * k_ss;
-            vptr[i] += (long)KVBLK * v_ss;
         }
     };
     auto write_tile_lds = [&]() {
@@ -1171,10 +1156,15 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_v4_kernel(
         }
     };
 
-    // one tile body, specialized at compile time on tail-key masking:
-    // only the LAST tile pays the per-element bounds compare/select.
-    auto tile_body = [&](int kv0, auto mask_c) {
-        constexpr bool MASK = decltype(mask_c)::value;
+    issue_tile_loads(0);
+    const int n_tiles = (Sk + KVBLK - 1) / KVBLK;
+    for (int t = 0; t < n_tiles; ++t) {
+        const int kv0 = t * KVBLK;
+        __syncthreads();
+        write_tile_lds();
+        __syncthreads();
+        if (t + 1 < n_tiles) issue_tile_loads(kv0 + KVBLK);
+
         // ---- swapped QK^T: S^T[key][row] for two 32-key tiles ------------
         f32x16 st[2];
 #pragma unroll
@@ -1184,6 +1174,7 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_v4_kernel(
             __builtin_amdgcn_s_setprio(1);
 #pragma unroll
             for (int kk = 0; kk < KK; ++kk) {
+                // A = K chunk: lane holds K[kt*32 + l32][kk*16 + hi*8 + j]
                 bf16x8 afrag = *reinterpret_cast<const bf16x8*>(
                     &k_lds[(kt * 32 + l32) * KPAD + kk * 16 + hi * 8]);
                 st[kt] = mfma32x32x16(afrag, qfrag[kk], st[kt]);
@@ -1192,43 +1183,42 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_v4_kernel(
         }
 
         // ---- lane-local online softmax (this lane's row = q0 + l32) ------
+        // value (kt, reg) = S[row][key = kt*32 + (reg&3) + 8*(reg>>2) + 4*hi]
         float mx = -1e30f;
 #pragma unroll
         for (int kt = 0; kt < 2; ++kt)
 #pragma unroll
             for (int r = 0; r < 16; ++r) {
-                float sv = st[kt][r] * scale2;
-                if (MASK) {
-                    const int key =
-                        kv0 + kt * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
-                    sv = (key < Sk) ? sv : -1e30f;
-                }
+                const int key = kv0 + kt * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+                const float sv = (key < Sk) ? st[kt][r] * scale2 : -1e30f;
                 st[kt][r] = sv;
                 mx = fmaxf(mx, sv);
             }
-        mx = fmaxf(mx, __shfl_xor(mx, 32, 64));
+        mx = fmaxf(mx, __shfl_xor(mx, 32, 64));  // partner holds the row's other 32 keys
         const float mnew = fmaxf(m_run, mx);
-        const float alpha = __builtin_amdgcn_exp2f(m_run - mnew);
+        const float alpha = exp2f(m_run - mnew);
         m_run = mnew;
         float ps = 0.f;
 #pragma unroll
         for (int kt = 0; kt < 2; ++kt)
 #pragma unroll
             for (int r = 0; r < 16; ++r) {
-                const float pv_ = __builtin_amdgcn_exp2f(st[kt][r] - mnew);
+                const float pv_ = exp2f(st[kt][r] - mnew);
                 st[kt][r] = pv_;
                 ps += pv_;
             }
         ps += __shfl_xor(ps, 32, 64);
         l_run = l_run * alpha + ps;
-        if (!__all(alpha == 1.f)) {  // wave-uniform: skip only when every
-#pragma unroll                       // lane's row max held (alpha==1 exact)
+        if (alpha != 1.f) {
+#pragma unroll
             for (int n = 0; n < NV; ++n)
 #pragma unroll
                 for (int r = 0; r < 16; ++r) o_acc[n][r] *= alpha;
         }
 
         // ---- P f32 -> bf16 fragments via cvt_pk + permlane32_swap --------
+        // chunk c (16 keys) uses regs 8*(c&1)..8*(c&1)+7 of st[c>>1]; after
+        // the half-swap each lane holds P[row l32][chunk base + hi*8 + j].
         bf16x8 pfrag[4];
 #pragma unroll
         for (int c = 0; c < 4; ++c) {
@@ -1251,6 +1241,7 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_v4_kernel(
         for (int c = 0; c < 4; ++c) {
 #pragma unroll
             for (int n = 0; n < NV; ++n) {
+                // A = V^T: lane holds V^T[n*32 + l32][16c + hi*8 + j]
                 const int dim = n * 32 + l32;
                 const int gsw = ((2 * c + hi) ^ ((dim >> 3) & 7)) << 3;
                 bf16x8 va = *reinterpret_cast<const bf16x8*>(
@@ -1259,21 +1250,6 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_v4_kernel(
             }
         }
         __builtin_amdgcn_s_setprio(0);
-    };
-
-    const int n_tiles = (Sk + KVBLK - 1) / KVBLK;
-    const int n_full = Sk / KVBLK;  // tiles with every key in range
-    issue_tile_loads(0, /*guard=*/0 >= n_full);
-    for (int t = 0; t < n_tiles; ++t) {
-        __syncthreads();
-        write_tile_lds();
-        __syncthreads();
-        if (t + 1 < n_tiles) issue_tile_loads((t + 1) * KVBLK, t + 1 >= n_full);
-        if (t < n_full) {
-            tile_body(t * KVBLK, std::integral_constant<bool, false>{});
-        } else {
-            tile_body(t * KVBLK, std::integral_constant<bool, true>{});
-        }
     }
 
     // ---- epilogue: O = O^T / l, row is lane-local -------------------------
