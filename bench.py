@@ -49,7 +49,9 @@ def parse_args():
                     choices=sorted(MODELS), help="model family")
     ap.add_argument("--batch", type=int, default=8)
     ap.add_argument("--px", type=int, default=1024)
-    ap.add_argument("--dtype", default="bf16", choices=["bf16", "fp16", "fp32"])
+    ap.add_argument("--dtype", default="bf16",
+                    choices=["bf16", "fp16", "fp32", "fp8"],
+                    help="fp8 = bf16 activations + e4m3fn GEMM weights (opt-in; NOT the headline dtype)")
     ap.add_argument("--tiny", action="store_true",
                     help="tiny config (CPU debugging only)")
     ap.add_argument("--no-graph", dest="graph", action="store_false",
@@ -58,7 +60,8 @@ def parse_args():
     return ap.parse_args()
 
 
-DTYPES = {"bf16": torch.bfloat16, "fp16": torch.float16, "fp32": torch.float32}
+DTYPES = {"bf16": torch.bfloat16, "fp16": torch.float16,
+          "fp32": torch.float32, "fp8": torch.bfloat16}
 
 
 def main():
@@ -80,6 +83,10 @@ def main():
     model = make(dev=dev, dtype=dtype, tiny=tiny)
     # replicate(): rank0's weights to every replica, flat bucketed RCCL bcast
     broadcast_module(model, src_rank=0)
+    if args.dtype == "fp8":
+        from comfyui_parallelanything_amd.models.quant import quantize_fp8
+
+        quantize_fp8(model)
 
     sizes = compute_split_sizes(args.batch, [1.0 / n] * n)
     if args.model == "wan":

@@ -1,0 +1,98 @@
+"""FP8 (OCP e4m3fn) serving mode for gfx950.
+
+MI355X runs fp8 MFMA at ~2x the bf16 rate (~5 PF dense). This module swaps
+large nn.Linear layers for FP8Linear: weights quantized once to e4m3fn with
+a per-tensor scale, activations quantized dynamically per call, matmul via
+torch._scaled_mm (hipBLASLt fp8 path). Norms/attention stay bf16.
+
+This is an OPT-IN mode (bench --dtype fp8; quantize_fp8(model)). The
+flagship benchmark stays bf16 — reduced-precision numbers are reported
+separately, never as the headline (bench contract).
+"""
+from __future__ import annotations
+
+import logging
+
+import torch
+from torch import nn
+
+log = logging.getLogger("parallelanything")
+
+FP8 = torch.float8_e4m3fn
+FP8_MAX = 448.0
+
+
+def _supports_scaled_mm() -> bool:
+    if not torch.cuda.is_available():
+        return False
+    try:
+        a = torch.randn(16, 16, device="cuda").to(FP8)
+        b = torch.randn(16, 16, device="cuda").to(FP8).t().contiguous().t()
+        s = torch.ones((), device="cuda")
+        torch._scaled_mm(a, b, scale_a=s, scale_b=s, out_dtype=torch.bfloat16)
+        return True
+    except Exception as err:  # noqa: BLE001
+        log.warning("fp8 _scaled_mm unavailable: %r", err)
+        return False
+
+
+class FP8Linear(nn.Module):
+    """Linear with e4m3fn weights + per-tensor scales (dynamic act quant)."""
+
+    def __init__(self, weight_fp8: torch.Tensor, w_scale: torch.Tensor,
+                 bias: torch.Tensor | None, out_dtype: torch.dtype):
+        super().__init__()
+        # weight stored [in, out] column-major-for-B as _scaled_mm wants
+        self.register_buffer("weight_fp8", weight_fp8, persistent=True)
+        self.register_buffer("w_scale", w_scale, persistent=True)
+        if bias is not None:
+            self.register_buffer("bias", bias, persistent=True)
+        else:
+            self.bias = None
+        self.out_dtype = out_dtype
+
+    @classmethod
+    def from_linear(cls, lin: nn.Linear) -> "FP8Linear":
+        w = lin.weight.detach().float()  # [out, in]
+        w_scale = (w.abs().amax() / FP8_MAX).clamp(min=1e-12)
+        w8 = (w / w_scale).clamp(-FP8_MAX, FP8_MAX).to(FP8)
+        # _scaled_mm wants B as [K, N] column-major: pass w8.t() (a view of
+        # the row-major [out, in] weight IS column-major [in, out])
+        bias = lin.bias.detach().clone() if lin.bias is not None else None
+        return cls(w8, w_scale.to(w.device), bias, lin.weight.dtype)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        shape = x.shape
+        x2 = x.reshape(-1, shape[-1])
+        x_scale = (x2.abs().amax().float() / FP8_MAX).clamp(min=1e-12)
+        x8 = (x2.float() / x_scale).clamp(-FP8_MAX, FP8_MAX).to(FP8)
+        y = torch._scaled_mm(
+            x8, self.weight_fp8.t(),
+            scale_a=x_scale, scale_b=self.w_scale,
+            bias=self.bias.to(self.out_dtype) if self.bias is not None else None,
+            out_dtype=self.out_dtype,
+        )
+        return y.reshape(*shape[:-1], y.shape[-1])
+
+
+def quantize_fp8(model: nn.Module, min_features: int = 1024) -> int:
+    """Swap large Linears for FP8Linear in place; returns the swap count.
+
+    Small projections (timestep/vector embedders, modulation heads with
+    tiny batch) stay bf16 — fp8 pays off on the token-major GEMMs.
+    """
+    if not _supports_scaled_mm():
+        raise RuntimeError(
+            "fp8 mode needs torch._scaled_mm on a gfx950 device"
+        )
+    n = 0
+    for parent in model.modules():
+        for name, child in list(parent.named_children()):
+            if isinstance(child, nn.Linear) and (
+                child.in_features >= min_features
+                and child.out_features >= min_features
+            ):
+                setattr(parent, name, FP8Linear.from_linear(child))
+                n += 1
+    log.info("fp8: quantized %d Linear layers", n)
+    return n
