@@ -86,3 +86,16 @@ def test_vram_balancer_reads_hbm():
 
     free = get_free_vram_mb("cuda:0")
     assert free > 10_000, f"expected >10 GB free HBM, got {free} MB"
+
+
+@pytest.mark.parametrize("name", ["sdxl", "zimage", "wan"])
+def test_model_families_gpu_tiny(name):
+    from comfyui_parallelanything_amd.models.registry import MODELS
+
+    make, inputs = MODELS[name]
+    m = make(dev="cuda:0", dtype=torch.bfloat16, tiny=True)
+    x, t, c, kw = inputs(2, dev="cuda:0", dtype=torch.bfloat16, tiny=True)
+    with torch.no_grad():
+        out = m(x, t, context=c, **kw)
+    assert out.shape == x.shape
+    assert torch.isfinite(out.float()).all()

@@ -151,3 +151,12 @@ def test_lora_detection_forces_lead_copy():
     eng = dm._parallel_engine
     assert eng.replicas["cpu"] is not dm
     cleanup_parallel_model(dm)
+
+
+def test_cli_tiny_runs(capsys):
+    from comfyui_parallelanything_amd.cli import main
+
+    main(["--model", "sd15", "--devices", "cpu,cpu", "--batch", "2",
+          "--steps", "2", "--tiny", "--no-balance"])
+    out = capsys.readouterr().out
+    assert "images_per_s" in out
