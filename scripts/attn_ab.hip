@@ -1,0 +1,351 @@
+// Standalone within-probe A/B harness for the v4 attention kernel (gfx950).
+// Variants share one process and interleave rounds (guide §5.4 rule 24) so
+// deltas are reliable despite cross-run/DVFS noise.
+//
+// Variants:
+//   0: shipped v4 (baseline, mirrors pa_ops.hip)
+//   1: + raw v_exp_f32 (__builtin_amdgcn_exp2f) in the softmax
+//   2: + per-thread incremented source pointers (no per-tile 64-bit math)
+//   3: + compile-time tail-mask specialization (mask only the last tile)
+//   4: 1+2+3 combined
+//
+// Build: hipcc --offload-arch=gfx950 -O3 -std=c++17 attn_ab.hip -o attn_ab
+// Run:   ./attn_ab [rounds]
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <cstdio>
+#include <cstdlib>
+#include <vector>
+#include <algorithm>
+#include <cmath>
+
+using bf16 = __hip_bfloat16;
+typedef __attribute__((ext_vector_type(8))) short bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+typedef __attribute__((ext_vector_type(16))) float f32x16;
+#define PA_DEV __device__ __forceinline__
+#define PA_LOG2E 1.4426950408889634f
+
+PA_DEV float bf2f(bf16 v) { return __bfloat162float(v); }
+PA_DEV bf16 f2bf(float v) { return __float2bfloat16(v); }
+PA_DEV f32x16 mfma32x32x16(bf16x8 a, bf16x8 b, f32x16 c) {
+    return __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, c, 0, 0, 0);
+}
+PA_DEV unsigned int cvt_pk_bf16(float lo, float hi) {
+    unsigned int r;
+    asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(r) : "v"(lo), "v"(hi));
+    return r;
+}
+
+template <int D, int VAR>
+__global__ __launch_bounds__(256, 2) void attn_ab_kernel(
+    const bf16* __restrict__ q, const bf16* __restrict__ k,
+    const bf16* __restrict__ v, bf16* __restrict__ out,
+    int S, int Sk, float scale, int H) {
+    constexpr int KVBLK = 64;
+    constexpr int WAVES = 4;
+    constexpr int THREADS = WAVES * 64;
+    constexpr int KPAD = D + 8;
+    constexpr int VPAD = KVBLK + 8;
+    constexpr int KK = D / 16;
+    constexpr int NV = D / 32;
+    constexpr int KVECS = (KVBLK * D) / (8 * THREADS);
+    constexpr bool RAW_EXP = (VAR == 1 || VAR == 4);
+    constexpr bool PTR_INC = (VAR == 2 || VAR == 4);
+    constexpr bool TAIL_SPEC = (VAR == 3 || VAR == 4);
+
+    __shared__ bf16 k_lds[KVBLK * KPAD];
+    __shared__ bf16 v_lds[D * VPAD];
+
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wid = tid >> 6;
+    const int l32 = lane & 31;
+    const int hi = lane >> 5;
+
+    // XCD-affine decode (1-D grid)
+    const int nq = (S + WAVES * 32 - 1) / (WAVES * 32);
+    const long id = blockIdx.x;
+    const long bh = (id & 7) + 8 * ((id >> 3) / nq);
+    const int qtile = (int)((id >> 3) % nq);
+    const long b = bh / H;
+    const int h = (int)(bh % H);
+    const int q0 = qtile * (WAVES * 32) + wid * 32;
+    const int ss = H * D;  // packed BSHD row stride
+
+    const bf16* qp = q + (b * (long)S + 0) * ss + (long)h * D;
+    const bf16* kp = k + (b * (long)Sk + 0) * ss + (long)h * D;
+    const bf16* vp = v + (b * (long)Sk + 0) * ss + (long)h * D;
+    bf16* op = out + (b * (long)S + 0) * ss + (long)h * D;
+
+    bf16x8 qfrag[KK];
+    {
+        const int row = q0 + l32;
+        const int rr = row < S ? row : S - 1;
+#pragma unroll
+        for (int kk = 0; kk < KK; ++kk)
+            qfrag[kk] = *reinterpret_cast<const bf16x8*>(
+                qp + (long)rr * ss + kk * 16 + hi * 8);
+    }
+
+    f32x16 o_acc[NV];
+#pragma unroll
+    for (int n = 0; n < NV; ++n)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) o_acc[n][r] = 0.f;
+    float m_run = -1e30f, l_run = 0.f;
+    const float scale2 = scale * PA_LOG2E;
+
+    bf16x8 kreg[KVECS], vreg[KVECS];
+    const bf16* kptr[KVECS];
+    const bf16* vptr[KVECS];
+    int lrow[KVECS];
+    if (PTR_INC) {
+#pragma unroll
+        for (int i = 0; i < KVECS; ++i) {
+            const int idx = tid + i * THREADS;
+            const int row = idx / (D / 8);
+            const int col = (idx % (D / 8)) * 8;
+            lrow[i] = row;
+            kptr[i] = kp + (long)row * ss + col;
+            vptr[i] = vp + (long)row * ss + col;
+        }
+    }
+
+    auto issue_tile_loads = [&](int kv0) {
+#pragma unroll
+        for (int i = 0; i < KVECS; ++i) {
+            if (PTR_INC) {
+                if (kv0 + lrow[i] < Sk) {
+                    kreg[i] = *reinterpret_cast<const bf16x8*>(kptr[i]);
+                    vreg[i] = *reinterpret_cast<const bf16x8*>(vptr[i]);
+                } else {
+                    kreg[i] = bf16x8{0,0,0,0,0,0,0,0};
+                    vreg[i] = bf16x8{0,0,0,0,0,0,0,0};
+                }
+                kptr[i] += (long)KVBLK * ss;
+                vptr[i] += (long)KVBLK * ss;
+            } else {
+                const int idx = tid + i * THREADS;
+                const int row = idx / (D / 8);
+                const int col = (idx % (D / 8)) * 8;
+                const int src = kv0 + row;
+                if (src < Sk) {
+                    kreg[i] = *reinterpret_cast<const bf16x8*>(kp + (long)src * ss + col);
+                    vreg[i] = *reinterpret_cast<const bf16x8*>(vp + (long)src * ss + col);
+                } else {
+                    kreg[i] = bf16x8{0,0,0,0,0,0,0,0};
+                    vreg[i] = bf16x8{0,0,0,0,0,0,0,0};
+                }
+            }
+        }
+    };
+    auto write_tile_lds = [&]() {
+#pragma unroll
+        for (int i = 0; i < KVECS; ++i) {
+            const int idx = tid + i * THREADS;
+            const int row = idx / (D / 8);
+            const int col = (idx % (D / 8)) * 8;
+            *reinterpret_cast<bf16x8*>(&k_lds[row * KPAD + col]) = kreg[i];
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                const int dim = col + j;
+                const int key_swz =
+                    (((row >> 3) ^ ((dim >> 3) & 7)) << 3) | (row & 7);
+                v_lds[dim * VPAD + key_swz] =
+                    __ushort_as_bfloat16((unsigned short)vreg[i][j]);
+            }
+        }
+    };
+
+    auto exp2x = [](float x) {
+        if (RAW_EXP) return __builtin_amdgcn_exp2f(x);
+        return exp2f(x);
+    };
+
+    auto tile = [&](int kv0, bool mask) {
+        f32x16 st[2];
+#pragma unroll
+        for (int kt = 0; kt < 2; ++kt) {
+#pragma unroll
+            for (int r = 0; r < 16; ++r) st[kt][r] = 0.f;
+            __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+            for (int kk = 0; kk < KK; ++kk) {
+                bf16x8 afrag = *reinterpret_cast<const bf16x8*>(
+                    &k_lds[(kt * 32 + l32) * KPAD + kk * 16 + hi * 8]);
+                st[kt] = mfma32x32x16(afrag, qfrag[kk], st[kt]);
+            }
+            __builtin_amdgcn_s_setprio(0);
+        }
+        float mx = -1e30f;
+#pragma unroll
+        for (int kt = 0; kt < 2; ++kt)
+#pragma unroll
+            for (int r = 0; r < 16; ++r) {
+                float sv = st[kt][r] * scale2;
+                if (mask) {
+                    const int key =
+                        kv0 + kt * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+                    sv = (key < Sk) ? sv : -1e30f;
+                }
+                st[kt][r] = sv;
+                mx = fmaxf(mx, sv);
+            }
+        mx = fmaxf(mx, __shfl_xor(mx, 32, 64));
+        const float mnew = fmaxf(m_run, mx);
+        const float alpha = exp2x(m_run - mnew);
+        m_run = mnew;
+        float ps = 0.f;
+#pragma unroll
+        for (int kt = 0; kt < 2; ++kt)
+#pragma unroll
+            for (int r = 0; r < 16; ++r) {
+                const float pv_ = exp2x(st[kt][r] - mnew);
+                st[kt][r] = pv_;
+                ps += pv_;
+            }
+        ps += __shfl_xor(ps, 32, 64);
+        l_run = l_run * alpha + ps;
+        if (alpha != 1.f) {
+#pragma unroll
+            for (int n = 0; n < NV; ++n)
+#pragma unroll
+                for (int r = 0; r < 16; ++r) o_acc[n][r] *= alpha;
+        }
+        bf16x8 pfrag[4];
+#pragma unroll
+        for (int c = 0; c < 4; ++c) {
+            const f32x16& sv = st[c >> 1];
+            const int rb = 8 * (c & 1);
+            unsigned int w0 = cvt_pk_bf16(sv[rb + 0], sv[rb + 1]);
+            unsigned int w1 = cvt_pk_bf16(sv[rb + 2], sv[rb + 3]);
+            unsigned int w2 = cvt_pk_bf16(sv[rb + 4], sv[rb + 5]);
+            unsigned int w3 = cvt_pk_bf16(sv[rb + 6], sv[rb + 7]);
+            auto r02 = __builtin_amdgcn_permlane32_swap(w0, w2, false, false);
+            auto r13 = __builtin_amdgcn_permlane32_swap(w1, w3, false, false);
+            unsigned int d[4] = {(unsigned int)r02[0], (unsigned int)r13[0],
+                                 (unsigned int)r02[1], (unsigned int)r13[1]};
+            pfrag[c] = *reinterpret_cast<bf16x8*>(d);
+        }
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int c = 0; c < 4; ++c) {
+#pragma unroll
+            for (int n = 0; n < NV; ++n) {
+                const int dim = n * 32 + l32;
+                const int gsw = ((2 * c + hi) ^ ((dim >> 3) & 7)) << 3;
+                bf16x8 va = *reinterpret_cast<const bf16x8*>(
+                    &v_lds[dim * VPAD + gsw]);
+                o_acc[n] = mfma32x32x16(va, pfrag[c], o_acc[n]);
+            }
+        }
+        __builtin_amdgcn_s_setprio(0);
+    };
+
+    const int n_tiles = (Sk + KVBLK - 1) / KVBLK;
+    const int n_full = Sk / KVBLK;
+    issue_tile_loads(0);
+    for (int t = 0; t < n_tiles; ++t) {
+        __syncthreads();
+        write_tile_lds();
+        __syncthreads();
+        if (t + 1 < n_tiles) issue_tile_loads((t + 1) * KVBLK);
+        tile(t * KVBLK, !TAIL_SPEC || t >= n_full);
+    }
+
+    const int row = q0 + l32;
+    if (row < S) {
+        const float inv_l = (l_run > 0.f) ? 1.f / l_run : 0.f;
+#pragma unroll
+        for (int n = 0; n < NV; ++n)
+#pragma unroll
+            for (int r2 = 0; r2 < 4; ++r2) {
+                const int dim0 = n * 32 + 8 * r2 + 4 * hi;
+                unsigned short pack[4];
+#pragma unroll
+                for (int j = 0; j < 4; ++j)
+                    pack[j] = __bfloat16_as_ushort(
+                        f2bf(o_acc[n][r2 * 4 + j] * inv_l));
+                *reinterpret_cast<unsigned long long*>(
+                    op + (long)row * ss + dim0) =
+                    *reinterpret_cast<unsigned long long*>(pack);
+            }
+    }
+}
+
+#define HIP_CHECK(x) do { hipError_t e = (x); if (e) { \
+    printf("HIP error %d at %d\n", e, __LINE__); exit(1); } } while (0)
+
+int main(int argc, char** argv) {
+    const int rounds = argc > 1 ? atoi(argv[1]) : 8;
+    const int B = 8, H = 24, S = 4608, D = 128;
+    const long n = (long)B * S * H * D;
+    bf16 *q, *k, *v, *o;
+    HIP_CHECK(hipMalloc(&q, n * 2));
+    HIP_CHECK(hipMalloc(&k, n * 2));
+    HIP_CHECK(hipMalloc(&v, n * 2));
+    HIP_CHECK(hipMalloc(&o, n * 2));
+    // random-ish fill on host (rule 25: never bench zero-filled)
+    {
+        std::vector<unsigned short> h(1 << 20);
+        unsigned x = 12345;
+        for (auto& e : h) {
+            x = x * 1664525u + 1013904223u;
+            float f = ((x >> 8) / 8388608.0f) * 2.f - 1.f;
+            unsigned int bits; __builtin_memcpy(&bits, &f, 4);
+            e = (unsigned short)(bits >> 16);  // truncate-to-bf16 on host
+        }
+        for (long off = 0; off < n; off += (1 << 20)) {
+            long len = std::min<long>(1 << 20, n - off);
+            HIP_CHECK(hipMemcpy(q + off, h.data(), len * 2, hipMemcpyHostToDevice));
+            HIP_CHECK(hipMemcpy(k + off, h.data(), len * 2, hipMemcpyHostToDevice));
+            HIP_CHECK(hipMemcpy(v + off, h.data(), len * 2, hipMemcpyHostToDevice));
+        }
+    }
+    const float scale = 1.0f / sqrtf((float)D);
+    const int nq = (S + 127) / 128;
+    dim3 grid(nq * B * H), blk(256);
+    const double tf = 4.0 * B * H * (double)S * S * D / 1e12;
+
+    auto run = [&](int var) {
+        switch (var) {
+        case 0: hipLaunchKernelGGL((attn_ab_kernel<128, 0>), grid, blk, 0, 0, q, k, v, o, S, S, scale, H); break;
+        case 1: hipLaunchKernelGGL((attn_ab_kernel<128, 1>), grid, blk, 0, 0, q, k, v, o, S, S, scale, H); break;
+        case 2: hipLaunchKernelGGL((attn_ab_kernel<128, 2>), grid, blk, 0, 0, q, k, v, o, S, S, scale, H); break;
+        case 3: hipLaunchKernelGGL((attn_ab_kernel<128, 3>), grid, blk, 0, 0, q, k, v, o, S, S, scale, H); break;
+        case 4: hipLaunchKernelGGL((attn_ab_kernel<128, 4>), grid, blk, 0, 0, q, k, v, o, S, S, scale, H); break;
+        }
+    };
+
+    // warmup
+    for (int var = 0; var < 5; ++var) run(var);
+    HIP_CHECK(hipDeviceSynchronize());
+
+    double best[5] = {1e30, 1e30, 1e30, 1e30, 1e30}, sum[5] = {};
+    for (int r = 0; r < rounds; ++r) {
+        for (int var = 0; var < 5; ++var) {
+            hipEvent_t e0, e1;
+            HIP_CHECK(hipEventCreate(&e0));
+            HIP_CHECK(hipEventCreate(&e1));
+            HIP_CHECK(hipEventRecord(e0));
+            for (int it = 0; it < 3; ++it) run(var);
+            HIP_CHECK(hipEventRecord(e1));
+            HIP_CHECK(hipEventSynchronize(e1));
+            float ms;
+            HIP_CHECK(hipEventElapsedTime(&ms, e0, e1));
+            double per = ms / 3.0;
+            if (per < best[var]) best[var] = per;
+            sum[var] += per;
+            hipEventDestroy(e0);
+            hipEventDestroy(e1);
+        }
+    }
+    const char* names[5] = {"baseline", "raw_exp2", "ptr_inc", "tail_spec", "all"};
+    for (int var = 0; var < 5; ++var)
+        printf("%-10s best %7.3f ms (%6.1f TF/s)  mean %7.3f ms\n",
+               names[var], best[var], tf / (best[var] / 1e3),
+               sum[var] / rounds);
+    return 0;
+}
