@@ -1196,14 +1196,14 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_v4_kernel(
             }
         mx = fmaxf(mx, __shfl_xor(mx, 32, 64));  // partner holds the row's other 32 keys
         const float mnew = fmaxf(m_run, mx);
-        const float alpha = exp2f(m_run - mnew);
+        const float alpha = __builtin_amdgcn_exp2f(m_run - mnew);
         m_run = mnew;
         float ps = 0.f;
 #pragma unroll
         for (int kt = 0; kt < 2; ++kt)
 #pragma unroll
             for (int r = 0; r < 16; ++r) {
-                const float pv_ = exp2f(st[kt][r] - mnew);
+                const float pv_ = __builtin_amdgcn_exp2f(st[kt][r] - mnew);
                 st[kt][r] = pv_;
                 ps += pv_;
             }
