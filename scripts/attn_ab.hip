@@ -3,11 +3,11 @@
 // deltas are reliable despite cross-run/DVFS noise.
 //
 // Variants:
-//   0: shipped v4 (baseline, mirrors pa_ops.hip)
-//   1: + raw v_exp_f32 (__builtin_amdgcn_exp2f) in the softmax
-//   2: + per-thread incremented source pointers (no per-tile 64-bit math)
-//   3: + compile-time tail-mask specialization (mask only the last tile)
-//   4: 1+2+3 combined
+//   0: shipped v4 (baseline; raw v_exp_f32 softmax)
+//   1: + interleaved kt MFMA order in QK^T (alternate accumulators)
+//   2: + unconditional alpha rescale (no per-lane branch)
+//   3: + no s_setprio
+//   4: 0 with paired V^T writes (b32 via shfl_xor 16)
 //
 // Build: hipcc --offload-arch=gfx950 -O3 -std=c++17 attn_ab.hip -o attn_ab
 // Run:   ./attn_ab [rounds]
@@ -51,9 +51,13 @@ __global__ __launch_bounds__(256, 2) void attn_ab_kernel(
     constexpr int KK = D / 16;
     constexpr int NV = D / 32;
     constexpr int KVECS = (KVBLK * D) / (8 * THREADS);
-    constexpr bool RAW_EXP = (VAR == 1 || VAR == 4);
-    constexpr bool PTR_INC = (VAR == 2 || VAR == 4);
-    constexpr bool TAIL_SPEC = (VAR == 3 || VAR == 4);
+    constexpr bool RAW_EXP = true;
+    constexpr bool PTR_INC = false;
+    constexpr bool TAIL_SPEC = false;
+    constexpr bool KT_ILV = (VAR == 1);
+    constexpr bool UNCOND_ALPHA = (VAR == 2);
+    constexpr bool NO_PRIO = (VAR == 3);
+    constexpr bool VPAIR = (VAR == 4);
 
     __shared__ bf16 k_lds[KVBLK * KPAD];
     __shared__ bf16 v_lds[D * VPAD];
@@ -148,37 +152,70 @@ __global__ __launch_bounds__(256, 2) void attn_ab_kernel(
             const int row = idx / (D / 8);
             const int col = (idx % (D / 8)) * 8;
             *reinterpret_cast<bf16x8*>(&k_lds[row * KPAD + col]) = kreg[i];
+            if (VPAIR && D == 128) {
+                // pair keys (row, row^1) via shfl_xor(16): even-key threads
+                // write dims col..col+3 as b32, odd-key threads col+4..col+7
+                const bool even = ((lane >> 4) & 1) == 0;
 #pragma unroll
-            for (int j = 0; j < 8; ++j) {
-                const int dim = col + j;
-                const int key_swz =
-                    (((row >> 3) ^ ((dim >> 3) & 7)) << 3) | (row & 7);
-                v_lds[dim * VPAD + key_swz] =
-                    __ushort_as_bfloat16((unsigned short)vreg[i][j]);
+                for (int jj = 0; jj < 4; ++jj) {
+                    const int j = even ? jj : jj + 4;
+                    const unsigned short mine =
+                        (unsigned short)vreg[i][j];
+                    const unsigned short partner = (unsigned short)__shfl_xor(
+                        (int)(unsigned short)vreg[i][j ^ 4], 16, 64);
+                    const unsigned short lo_key = even ? mine : partner;
+                    const unsigned short hi_key = even ? partner : mine;
+                    const int dim = col + j;
+                    const int key0 = row & ~1;
+                    const int key_swz =
+                        (((key0 >> 3) ^ ((dim >> 3) & 7)) << 3) | (key0 & 7);
+                    *reinterpret_cast<unsigned int*>(
+                        &v_lds[dim * VPAD + key_swz]) =
+                        (unsigned int)lo_key | ((unsigned int)hi_key << 16);
+                }
+            } else {
+#pragma unroll
+                for (int j = 0; j < 8; ++j) {
+                    const int dim = col + j;
+                    const int key_swz =
+                        (((row >> 3) ^ ((dim >> 3) & 7)) << 3) | (row & 7);
+                    v_lds[dim * VPAD + key_swz] =
+                        __ushort_as_bfloat16((unsigned short)vreg[i][j]);
+                }
             }
         }
     };
 
-    auto exp2x = [](float x) {
-        if (RAW_EXP) return __builtin_amdgcn_exp2f(x);
-        return exp2f(x);
-    };
+    auto exp2x = [](float x) { return __builtin_amdgcn_exp2f(x); };
+    (void)RAW_EXP; (void)PTR_INC; (void)TAIL_SPEC;
 
     auto tile = [&](int kv0, bool mask) {
         f32x16 st[2];
 #pragma unroll
-        for (int kt = 0; kt < 2; ++kt) {
+        for (int kt = 0; kt < 2; ++kt)
 #pragma unroll
             for (int r = 0; r < 16; ++r) st[kt][r] = 0.f;
-            __builtin_amdgcn_s_setprio(1);
+        if (!NO_PRIO) __builtin_amdgcn_s_setprio(1);
+        if (KT_ILV) {
 #pragma unroll
-            for (int kk = 0; kk < KK; ++kk) {
-                bf16x8 afrag = *reinterpret_cast<const bf16x8*>(
-                    &k_lds[(kt * 32 + l32) * KPAD + kk * 16 + hi * 8]);
-                st[kt] = mfma32x32x16(afrag, qfrag[kk], st[kt]);
-            }
-            __builtin_amdgcn_s_setprio(0);
+            for (int kk = 0; kk < KK; ++kk)
+#pragma unroll
+                for (int kt = 0; kt < 2; ++kt) {
+                    bf16x8 afrag = *reinterpret_cast<const bf16x8*>(
+                        &k_lds[(kt * 32 + l32) * KPAD + kk * 16 + hi * 8]);
+                    st[kt] = mfma32x32x16(afrag, qfrag[kk], st[kt]);
+                }
+        } else {
+#pragma unroll
+            for (int kt = 0; kt < 2; ++kt)
+#pragma unroll
+                for (int kk = 0; kk < KK; ++kk) {
+                    bf16x8 afrag = *reinterpret_cast<const bf16x8*>(
+                        &k_lds[(kt * 32 + l32) * KPAD + kk * 16 + hi * 8]);
+                    st[kt] = mfma32x32x16(afrag, qfrag[kk], st[kt]);
+                }
         }
+        if (!NO_PRIO) __builtin_amdgcn_s_setprio(0);
         float mx = -1e30f;
 #pragma unroll
         for (int kt = 0; kt < 2; ++kt)
@@ -208,7 +245,12 @@ __global__ __launch_bounds__(256, 2) void attn_ab_kernel(
             }
         ps += __shfl_xor(ps, 32, 64);
         l_run = l_run * alpha + ps;
-        if (alpha != 1.f) {
+        if (UNCOND_ALPHA) {
+#pragma unroll
+            for (int n = 0; n < NV; ++n)
+#pragma unroll
+                for (int r = 0; r < 16; ++r) o_acc[n][r] *= alpha;
+        } else if (alpha != 1.f) {
 #pragma unroll
             for (int n = 0; n < NV; ++n)
 #pragma unroll
@@ -229,7 +271,7 @@ __global__ __launch_bounds__(256, 2) void attn_ab_kernel(
                                  (unsigned int)r02[1], (unsigned int)r13[1]};
             pfrag[c] = *reinterpret_cast<bf16x8*>(d);
         }
-        __builtin_amdgcn_s_setprio(1);
+        if (!NO_PRIO) __builtin_amdgcn_s_setprio(1);
 #pragma unroll
         for (int c = 0; c < 4; ++c) {
 #pragma unroll
@@ -241,7 +283,7 @@ __global__ __launch_bounds__(256, 2) void attn_ab_kernel(
                 o_acc[n] = mfma32x32x16(va, pfrag[c], o_acc[n]);
             }
         }
-        __builtin_amdgcn_s_setprio(0);
+        if (!NO_PRIO) __builtin_amdgcn_s_setprio(0);
     };
 
     const int n_tiles = (Sk + KVBLK - 1) / KVBLK;
