@@ -3,11 +3,9 @@
 // deltas are reliable despite cross-run/DVFS noise.
 //
 // Variants:
-//   0: shipped v4 (baseline; raw v_exp_f32 softmax)
-//   1: + interleaved kt MFMA order in QK^T (alternate accumulators)
-//   2: + unconditional alpha rescale (no per-lane branch)
-//   3: + no s_setprio
-//   4: 0 with paired V^T writes (b32 via shfl_xor 16)
+//   0: shipped v4 (4-wave blocks, 128 q-rows; 2 blocks/CU desynced)
+//   1: 8-wave blocks (256 q-rows; HALVES K/V re-streaming per (b,h))
+//   (both run at a FLUX shape and a WAN-like long-S shape)
 //
 // Build: hipcc --offload-arch=gfx950 -O3 -std=c++17 attn_ab.hip -o attn_ab
 // Run:   ./attn_ab [rounds]
@@ -44,7 +42,7 @@ __global__ __launch_bounds__(256, 2) void attn_ab_kernel(
     const bf16* __restrict__ v, bf16* __restrict__ out,
     int S, int Sk, float scale, int H) {
     constexpr int KVBLK = 64;
-    constexpr int WAVES = 4;
+    constexpr int WAVES = (VAR == 1) ? 8 : 4;
     constexpr int THREADS = WAVES * 64;
     constexpr int KPAD = D + 8;
     constexpr int VPAD = KVBLK + 8;
@@ -54,10 +52,10 @@ __global__ __launch_bounds__(256, 2) void attn_ab_kernel(
     constexpr bool RAW_EXP = true;
     constexpr bool PTR_INC = false;
     constexpr bool TAIL_SPEC = false;
-    constexpr bool KT_ILV = (VAR == 1);
-    constexpr bool UNCOND_ALPHA = (VAR == 2);
-    constexpr bool NO_PRIO = (VAR == 3);
-    constexpr bool VPAIR = (VAR == 4);
+    constexpr bool KT_ILV = false;
+    constexpr bool UNCOND_ALPHA = false;
+    constexpr bool NO_PRIO = false;
+    constexpr bool VPAIR = false;
 
     __shared__ bf16 k_lds[KVBLK * KPAD];
     __shared__ bf16 v_lds[D * VPAD];
@@ -322,72 +320,66 @@ __global__ __launch_bounds__(256, 2) void attn_ab_kernel(
 
 int main(int argc, char** argv) {
     const int rounds = argc > 1 ? atoi(argv[1]) : 8;
-    const int B = 8, H = 24, S = 4608, D = 128;
-    const long n = (long)B * S * H * D;
-    bf16 *q, *k, *v, *o;
-    HIP_CHECK(hipMalloc(&q, n * 2));
-    HIP_CHECK(hipMalloc(&k, n * 2));
-    HIP_CHECK(hipMalloc(&v, n * 2));
-    HIP_CHECK(hipMalloc(&o, n * 2));
-    // random-ish fill on host (rule 25: never bench zero-filled)
-    {
-        std::vector<unsigned short> h(1 << 20);
-        unsigned x = 12345;
-        for (auto& e : h) {
-            x = x * 1664525u + 1013904223u;
-            float f = ((x >> 8) / 8388608.0f) * 2.f - 1.f;
-            unsigned int bits; __builtin_memcpy(&bits, &f, 4);
-            e = (unsigned short)(bits >> 16);  // truncate-to-bf16 on host
+    struct Shape { int B, H, S; const char* name; };
+    Shape shapes[2] = {{8, 24, 4608, "flux"}, {1, 8, 30720, "long"}};
+    for (auto& sh : shapes) {
+        const int B = sh.B, H = sh.H, S = sh.S, D = 128;
+        const long n = (long)B * S * H * D;
+        bf16 *q, *k, *v, *o;
+        HIP_CHECK(hipMalloc(&q, n * 2));
+        HIP_CHECK(hipMalloc(&k, n * 2));
+        HIP_CHECK(hipMalloc(&v, n * 2));
+        HIP_CHECK(hipMalloc(&o, n * 2));
+        {
+            std::vector<unsigned short> h(1 << 20);
+            unsigned x = 12345;
+            for (auto& e : h) {
+                x = x * 1664525u + 1013904223u;
+                float f = ((x >> 8) / 8388608.0f) * 2.f - 1.f;
+                unsigned int bits; __builtin_memcpy(&bits, &f, 4);
+                e = (unsigned short)(bits >> 16);
+            }
+            for (long off = 0; off < n; off += (1 << 20)) {
+                long len = std::min<long>(1 << 20, n - off);
+                HIP_CHECK(hipMemcpy(q + off, h.data(), len * 2, hipMemcpyHostToDevice));
+                HIP_CHECK(hipMemcpy(k + off, h.data(), len * 2, hipMemcpyHostToDevice));
+                HIP_CHECK(hipMemcpy(v + off, h.data(), len * 2, hipMemcpyHostToDevice));
+            }
         }
-        for (long off = 0; off < n; off += (1 << 20)) {
-            long len = std::min<long>(1 << 20, n - off);
-            HIP_CHECK(hipMemcpy(q + off, h.data(), len * 2, hipMemcpyHostToDevice));
-            HIP_CHECK(hipMemcpy(k + off, h.data(), len * 2, hipMemcpyHostToDevice));
-            HIP_CHECK(hipMemcpy(v + off, h.data(), len * 2, hipMemcpyHostToDevice));
-        }
+        const float scale = 1.0f / sqrtf((float)D);
+        const double tf = 4.0 * B * H * (double)S * S * D / 1e12;
+        auto run = [&](int var) {
+            if (var == 0) {
+                dim3 grid(((S + 127) / 128) * B * H), blk(256);
+                hipLaunchKernelGGL((attn_ab_kernel<128, 0>), grid, blk, 0, 0,
+                                   q, k, v, o, S, S, scale, H);
+            } else {
+                dim3 grid(((S + 255) / 256) * B * H), blk(512);
+                hipLaunchKernelGGL((attn_ab_kernel<128, 1>), grid, blk, 0, 0,
+                                   q, k, v, o, S, S, scale, H);
+            }
+        };
+        for (int var = 0; var < 2; ++var) run(var);
+        HIP_CHECK(hipDeviceSynchronize());
+        double best[2] = {1e30, 1e30};
+        for (int r = 0; r < rounds; ++r)
+            for (int var = 0; var < 2; ++var) {
+                hipEvent_t e0, e1;
+                HIP_CHECK(hipEventCreate(&e0));
+                HIP_CHECK(hipEventCreate(&e1));
+                HIP_CHECK(hipEventRecord(e0));
+                for (int it = 0; it < 3; ++it) run(var);
+                HIP_CHECK(hipEventRecord(e1));
+                HIP_CHECK(hipEventSynchronize(e1));
+                float ms;
+                HIP_CHECK(hipEventElapsedTime(&ms, e0, e1));
+                if (ms / 3.0 < best[var]) best[var] = ms / 3.0;
+                (void)hipEventDestroy(e0);
+                (void)hipEventDestroy(e1);
+            }
+        printf("%s  waves4 %7.3f ms (%6.1f TF)   waves8 %7.3f ms (%6.1f TF)\n",
+               sh.name, best[0], tf / best[0] * 1e3, best[1], tf / best[1] * 1e3);
+        (void)hipFree(q); (void)hipFree(k); (void)hipFree(v); (void)hipFree(o);
     }
-    const float scale = 1.0f / sqrtf((float)D);
-    const int nq = (S + 127) / 128;
-    dim3 grid(nq * B * H), blk(256);
-    const double tf = 4.0 * B * H * (double)S * S * D / 1e12;
-
-    auto run = [&](int var) {
-        switch (var) {
-        case 0: hipLaunchKernelGGL((attn_ab_kernel<128, 0>), grid, blk, 0, 0, q, k, v, o, S, S, scale, H); break;
-        case 1: hipLaunchKernelGGL((attn_ab_kernel<128, 1>), grid, blk, 0, 0, q, k, v, o, S, S, scale, H); break;
-        case 2: hipLaunchKernelGGL((attn_ab_kernel<128, 2>), grid, blk, 0, 0, q, k, v, o, S, S, scale, H); break;
-        case 3: hipLaunchKernelGGL((attn_ab_kernel<128, 3>), grid, blk, 0, 0, q, k, v, o, S, S, scale, H); break;
-        case 4: hipLaunchKernelGGL((attn_ab_kernel<128, 4>), grid, blk, 0, 0, q, k, v, o, S, S, scale, H); break;
-        }
-    };
-
-    // warmup
-    for (int var = 0; var < 5; ++var) run(var);
-    HIP_CHECK(hipDeviceSynchronize());
-
-    double best[5] = {1e30, 1e30, 1e30, 1e30, 1e30}, sum[5] = {};
-    for (int r = 0; r < rounds; ++r) {
-        for (int var = 0; var < 5; ++var) {
-            hipEvent_t e0, e1;
-            HIP_CHECK(hipEventCreate(&e0));
-            HIP_CHECK(hipEventCreate(&e1));
-            HIP_CHECK(hipEventRecord(e0));
-            for (int it = 0; it < 3; ++it) run(var);
-            HIP_CHECK(hipEventRecord(e1));
-            HIP_CHECK(hipEventSynchronize(e1));
-            float ms;
-            HIP_CHECK(hipEventElapsedTime(&ms, e0, e1));
-            double per = ms / 3.0;
-            if (per < best[var]) best[var] = per;
-            sum[var] += per;
-            hipEventDestroy(e0);
-            hipEventDestroy(e1);
-        }
-    }
-    const char* names[5] = {"baseline", "raw_exp2", "ptr_inc", "tail_spec", "all"};
-    for (int var = 0; var < 5; ++var)
-        printf("%-10s best %7.3f ms (%6.1f TF/s)  mean %7.3f ms\n",
-               names[var], best[var], tf / (best[var] / 1e3),
-               sum[var] / rounds);
     return 0;
 }
