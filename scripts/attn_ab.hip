@@ -37,7 +37,7 @@ PA_DEV unsigned int cvt_pk_bf16(float lo, float hi) {
 }
 
 template <int D, int VAR>
-__global__ __launch_bounds__(256, 2) void attn_ab_kernel(
+__global__ __launch_bounds__(VAR == 1 ? 512 : 256, 2) void attn_ab_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ k,
     const bf16* __restrict__ v, bf16* __restrict__ out,
     int S, int Sk, float scale, int H) {
@@ -359,7 +359,7 @@ int main(int argc, char** argv) {
                                    q, k, v, o, S, S, scale, H);
             }
         };
-        for (int var = 0; var < 2; ++var) run(var);
+        for (int var = 0; var < 2; ++var) { run(var); HIP_CHECK(hipGetLastError()); }
         HIP_CHECK(hipDeviceSynchronize());
         double best[2] = {1e30, 1e30};
         for (int r = 0; r < rounds; ++r)
