@@ -1050,7 +1050,7 @@ PA_DEV unsigned int cvt_pk_bf16(float lo, float hi) {
 }
 
 template <int D>
-__global__ __launch_bounds__(256, 2) void attn_fwd_v4_kernel(
+__global__ __launch_bounds__(512, 2) void attn_fwd_v4_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ k,
     const bf16* __restrict__ v, bf16* __restrict__ out,
     int S, int Sk, float scale, int H,
@@ -1059,7 +1059,7 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_v4_kernel(
     long v_bs, long v_hs, int v_ss,
     long o_bs, long o_hs, int o_ss, int xcd_grid) {
     constexpr int KVBLK = 64;
-    constexpr int WAVES = 4;
+    constexpr int WAVES = 8;
     constexpr int THREADS = WAVES * 64;
     constexpr int KPAD = D + 8;
     constexpr int VPAD = KVBLK + 8;
@@ -1375,7 +1375,7 @@ static at::Tensor attn_fwd_launch(at::Tensor q, at::Tensor k, at::Tensor v,
         return e && e[0] == '1';
     }();
     int xcd_grid = 0;
-    dim3 grid_v4((unsigned)((S + 127) / 128), (unsigned)((long)B * H));
+    dim3 grid_v4((unsigned)((S + 255) / 256), (unsigned)((long)B * H));
     dim3 grid((unsigned)((S + 255) / 256), (unsigned)((long)B * H));
 #define PA_ATTN_ARGS                                                          \
     (const bf16*)qc.data_ptr(), (const bf16*)kc.data_ptr(),                   \
@@ -1386,17 +1386,17 @@ static at::Tensor attn_fwd_launch(at::Tensor q, at::Tensor k, at::Tensor v,
     out.stride(b_ax), out.stride(h_ax), (int)out.stride(s_ax)
     if (!use_v3) {
         const long BH = (long)B * H;
-        const unsigned nq = (unsigned)((S + 127) / 128);
+        const unsigned nq = (unsigned)((S + 255) / 256);
         if (BH % 8 == 0) {
             // 1-D XCD-affine grid: same-(b,h) q-tiles share an XCD's L2
             grid_v4 = dim3((unsigned)(nq * BH), 1);
             xcd_grid = 1;
         }
         if (D == 128) {
-            hipLaunchKernelGGL(attn_fwd_v4_kernel<128>, grid_v4, dim3(256), 0,
+            hipLaunchKernelGGL(attn_fwd_v4_kernel<128>, grid_v4, dim3(512), 0,
                                cur_stream(), PA_ATTN_ARGS, xcd_grid);
         } else {
-            hipLaunchKernelGGL(attn_fwd_v4_kernel<64>, grid_v4, dim3(256), 0,
+            hipLaunchKernelGGL(attn_fwd_v4_kernel<64>, grid_v4, dim3(512), 0,
                                cur_stream(), PA_ATTN_ARGS, xcd_grid);
         }
     } else if (D == 128) {
