@@ -295,13 +295,18 @@ __global__ void gate_residual_bf16_kernel(const bf16* __restrict__ res,
 // Vectorized tanh-GELU (bf16 short8): the MLP activation between the two
 // hipBLASLt GEMMs ([B, S, 4*hidden] tensors; memory-bound).
 __global__ void gelu_tanh_bf16_kernel(const bf16* __restrict__ x,
-                                      bf16* __restrict__ out, long total8) {
+                                      bf16* __restrict__ out, long total8,
+                                      long rows, long w8, long in_stride8) {
+    // strided rows supported (last-dim slices of a fused projection read in
+    // place — no .contiguous() copy); output is written dense.
     const long stride = (long)gridDim.x * blockDim.x;
     const short8* xv = reinterpret_cast<const short8*>(x);
     short8* ov = reinterpret_cast<short8*>(out);
     for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total8;
          i += stride) {
-        short8 v = xv[i], o;
+        const long src = (in_stride8 == w8)
+            ? i : (i / w8) * in_stride8 + (i % w8);
+        short8 v = xv[src], o;
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
             const float f = bf2f(__ushort_as_bfloat16((unsigned short)v[j]));
@@ -1279,6 +1284,32 @@ __global__ __launch_bounds__(512, 2) void attn_fwd_v4_kernel(
 
 at::Tensor gelu_tanh(at::Tensor x) {
     CHECK_GPU(x);
+    // accept a 2-D-decomposable strided view: last dim contiguous, all
+    // leading dims collapsible to rows with ONE stride (e.g. a last-dim
+    // slice of a fused projection) — avoids a .contiguous() copy.
+    const int nd = x.dim();
+    bool rowable = x.stride(nd - 1) == 1;
+    long w = x.size(nd - 1), rows = 1, in_stride = x.stride(nd - 1) * w;
+    if (rowable && nd >= 2) {
+        in_stride = x.stride(nd - 2);
+        rows = x.numel() / w;
+        long expect = in_stride;
+        for (int d = nd - 3; d >= 0; --d) {
+            expect *= x.size(d + 1);
+            if (x.stride(d) != expect) { rowable = false; break; }
+        }
+    }
+    if (x.scalar_type() == at::kBFloat16 && rowable && (w % 8) == 0 &&
+        (in_stride % 8) == 0) {
+        auto out = at::empty(x.sizes(), x.options());
+        const long total8 = rows * (w / 8);
+        const int blocks = (int)std::min<long>((total8 + 255) / 256, 4096);
+        hipLaunchKernelGGL(gelu_tanh_bf16_kernel, dim3(blocks), dim3(256), 0,
+                           cur_stream(), (const bf16*)x.data_ptr(),
+                           (bf16*)out.data_ptr(), total8, rows, w / 8,
+                           in_stride / 8);
+        return out;
+    }
     auto xc = x.contiguous();
     auto out = at::empty_like(xc);
     if (xc.scalar_type() == at::kBFloat16 && (xc.numel() % 8) == 0) {
@@ -1286,7 +1317,8 @@ at::Tensor gelu_tanh(at::Tensor x) {
         const int blocks = (int)std::min<long>((total8 + 255) / 256, 4096);
         hipLaunchKernelGGL(gelu_tanh_bf16_kernel, dim3(blocks), dim3(256), 0,
                            cur_stream(), (const bf16*)xc.data_ptr(),
-                           (bf16*)out.data_ptr(), total8);
+                           (bf16*)out.data_ptr(), total8, xc.numel() / 8,
+                           xc.numel() / 8, xc.numel() / 8);
         return out;
     }
     return at::gelu(xc, "tanh");

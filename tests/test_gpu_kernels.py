@@ -265,3 +265,12 @@ def test_attn_single_head_batch():
     out = ops.attention(q, k, v)
     ref = R.attention(q.float(), k.float(), v.float())
     _cmp(out, ref, 2e-2, 2e-2, "attn BH=1")
+
+
+def test_gelu_tanh_strided_slice():
+    """gelu on a last-dim slice of a fused projection (no copy path)."""
+    proj = torch.randn(2, 64, 21504, device="cuda", dtype=torch.bfloat16)
+    mlp = proj[..., 9216:]
+    out = ops.gelu_tanh(mlp)
+    ref = torch.nn.functional.gelu(mlp.float(), approximate="tanh")
+    _cmp(out, ref, 2e-2, 2e-2, "gelu strided")
