@@ -3,9 +3,9 @@
 // deltas are reliable despite cross-run/DVFS noise.
 //
 // Variants:
-//   0: shipped v4 (4-wave blocks, 128 q-rows; 2 blocks/CU desynced)
-//   1: 8-wave blocks (256 q-rows; HALVES K/V re-streaming per (b,h))
-//   (both run at a FLUX shape and a WAN-like long-S shape)
+//   0: shipped v4 (8-wave blocks, KVBLK 64)
+//   1: 128-key supertiles: stage 128 keys per barrier pair, run two
+//      sequential 64-key softmax passes (half the barrier/stage rounds)
 //
 // Build: hipcc --offload-arch=gfx950 -O3 -std=c++17 attn_ab.hip -o attn_ab
 // Run:   ./attn_ab [rounds]
@@ -37,18 +37,19 @@ PA_DEV unsigned int cvt_pk_bf16(float lo, float hi) {
 }
 
 template <int D, int VAR>
-__global__ __launch_bounds__(VAR == 1 ? 512 : 256, 2) void attn_ab_kernel(
+__global__ __launch_bounds__(512, 2) void attn_ab_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ k,
     const bf16* __restrict__ v, bf16* __restrict__ out,
     int S, int Sk, float scale, int H) {
     constexpr int KVBLK = 64;
-    constexpr int WAVES = (VAR == 1) ? 8 : 4;
+    constexpr int WAVES = 8;
+    constexpr int SUPER = (VAR == 1) ? 2 : 1;
     constexpr int THREADS = WAVES * 64;
     constexpr int KPAD = D + 8;
     constexpr int VPAD = KVBLK + 8;
     constexpr int KK = D / 16;
     constexpr int NV = D / 32;
-    constexpr int KVECS = (KVBLK * D) / (8 * THREADS);
+    constexpr int KVECS = (SUPER * KVBLK * D) / (8 * THREADS);
     constexpr bool RAW_EXP = true;
     constexpr bool PTR_INC = false;
     constexpr bool TAIL_SPEC = false;
@@ -57,8 +58,8 @@ __global__ __launch_bounds__(VAR == 1 ? 512 : 256, 2) void attn_ab_kernel(
     constexpr bool NO_PRIO = false;
     constexpr bool VPAIR = false;
 
-    __shared__ bf16 k_lds[KVBLK * KPAD];
-    __shared__ bf16 v_lds[D * VPAD];
+    __shared__ bf16 k_lds[SUPER * KVBLK * KPAD];
+    __shared__ bf16 v_lds[SUPER * D * VPAD];
 
     const int tid = threadIdx.x;
     const int lane = tid & 63;
@@ -118,7 +119,7 @@ __global__ __launch_bounds__(VAR == 1 ? 512 : 256, 2) void attn_ab_kernel(
     auto issue_tile_loads = [&](int kv0) {
 #pragma unroll
         for (int i = 0; i < KVECS; ++i) {
-            if (PTR_INC) {
+            if (false) {
                 if (kv0 + lrow[i] < Sk) {
                     kreg[i] = *reinterpret_cast<const bf16x8*>(kptr[i]);
                     vreg[i] = *reinterpret_cast<const bf16x8*>(vptr[i]);
@@ -149,6 +150,8 @@ __global__ __launch_bounds__(VAR == 1 ? 512 : 256, 2) void attn_ab_kernel(
             const int idx = tid + i * THREADS;
             const int row = idx / (D / 8);
             const int col = (idx % (D / 8)) * 8;
+            const int sub = row / KVBLK;          // 64-key sub-tile id
+            const int srow = row % KVBLK;
             *reinterpret_cast<bf16x8*>(&k_lds[row * KPAD + col]) = kreg[i];
             if (VPAIR && D == 128) {
                 // pair keys (row, row^1) via shfl_xor(16): even-key threads
@@ -176,8 +179,8 @@ __global__ __launch_bounds__(VAR == 1 ? 512 : 256, 2) void attn_ab_kernel(
                 for (int j = 0; j < 8; ++j) {
                     const int dim = col + j;
                     const int key_swz =
-                        (((row >> 3) ^ ((dim >> 3) & 7)) << 3) | (row & 7);
-                    v_lds[dim * VPAD + key_swz] =
+                        (((srow >> 3) ^ ((dim >> 3) & 7)) << 3) | (srow & 7);
+                    v_lds[(sub * D + dim) * VPAD + key_swz] =
                         __ushort_as_bfloat16((unsigned short)vreg[i][j]);
                 }
             }
@@ -187,7 +190,7 @@ __global__ __launch_bounds__(VAR == 1 ? 512 : 256, 2) void attn_ab_kernel(
     auto exp2x = [](float x) { return __builtin_amdgcn_exp2f(x); };
     (void)RAW_EXP; (void)PTR_INC; (void)TAIL_SPEC;
 
-    auto tile = [&](int kv0, bool mask) {
+    auto tile = [&](int kv0, bool mask, int sub) {
         f32x16 st[2];
 #pragma unroll
         for (int kt = 0; kt < 2; ++kt)
@@ -277,22 +280,24 @@ __global__ __launch_bounds__(VAR == 1 ? 512 : 256, 2) void attn_ab_kernel(
                 const int dim = n * 32 + l32;
                 const int gsw = ((2 * c + hi) ^ ((dim >> 3) & 7)) << 3;
                 bf16x8 va = *reinterpret_cast<const bf16x8*>(
-                    &v_lds[dim * VPAD + gsw]);
+                    &v_lds[(sub * D + dim) * VPAD + gsw]);
                 o_acc[n] = mfma32x32x16(va, pfrag[c], o_acc[n]);
             }
         }
         if (!NO_PRIO) __builtin_amdgcn_s_setprio(0);
     };
 
-    const int n_tiles = (Sk + KVBLK - 1) / KVBLK;
-    const int n_full = Sk / KVBLK;
+    const int SBLK = SUPER * KVBLK;
+    const int n_super = (Sk + SBLK - 1) / SBLK;
     issue_tile_loads(0);
-    for (int t = 0; t < n_tiles; ++t) {
+    for (int t = 0; t < n_super; ++t) {
         __syncthreads();
         write_tile_lds();
         __syncthreads();
-        if (t + 1 < n_tiles) issue_tile_loads((t + 1) * KVBLK);
-        tile(t * KVBLK, !TAIL_SPEC || t >= n_full);
+        if (t + 1 < n_super) issue_tile_loads((t + 1) * SBLK);
+#pragma unroll
+        for (int sub = 0; sub < SUPER; ++sub)
+            tile(t * SBLK + sub * KVBLK, true, sub);
     }
 
     const int row = q0 + l32;
@@ -349,15 +354,13 @@ int main(int argc, char** argv) {
         const float scale = 1.0f / sqrtf((float)D);
         const double tf = 4.0 * B * H * (double)S * S * D / 1e12;
         auto run = [&](int var) {
-            if (var == 0) {
-                dim3 grid(((S + 127) / 128) * B * H), blk(256);
+            dim3 grid(((S + 255) / 256) * B * H), blk(512);
+            if (var == 0)
                 hipLaunchKernelGGL((attn_ab_kernel<128, 0>), grid, blk, 0, 0,
                                    q, k, v, o, S, S, scale, H);
-            } else {
-                dim3 grid(((S + 255) / 256) * B * H), blk(512);
+            else
                 hipLaunchKernelGGL((attn_ab_kernel<128, 1>), grid, blk, 0, 0,
                                    q, k, v, o, S, S, scale, H);
-            }
         };
         for (int var = 0; var < 2; ++var) { run(var); HIP_CHECK(hipGetLastError()); }
         HIP_CHECK(hipDeviceSynchronize());
@@ -377,7 +380,7 @@ int main(int argc, char** argv) {
                 (void)hipEventDestroy(e0);
                 (void)hipEventDestroy(e1);
             }
-        printf("%s  waves4 %7.3f ms (%6.1f TF)   waves8 %7.3f ms (%6.1f TF)\n",
+        printf("%s  kv64 %7.3f ms (%6.1f TF)   kv128 %7.3f ms (%6.1f TF)\n",
                sh.name, best[0], tf / best[0] * 1e3, best[1], tf / best[1] * 1e3);
         (void)hipFree(q); (void)hipFree(k); (void)hipFree(v); (void)hipFree(o);
     }
