@@ -98,3 +98,25 @@ def _comm_stats_worker(rank, world, tmpdir):
 
 def test_comm_stats(tmp_path):
     mp.spawn(_comm_stats_worker, args=(2, str(tmp_path)), nprocs=2, join=True)
+
+
+def _world4_worker(rank, world, tmpdir):
+    from comfyui_parallelanything_amd.parallel.dist import gatherv, scatterv
+
+    info = _init(rank, world, tmpdir)
+    sizes = [3, 2, 2, 1]  # weighted split, driver-style N=4
+    full = torch.arange(8 * 4, dtype=torch.float32).view(8, 4)
+    chunk = scatterv(full if rank == 0 else None, sizes, info, src=0,
+                     template=full)
+    assert chunk.shape[0] == sizes[rank]
+    out = gatherv(chunk + rank, sizes, info, dst=0)
+    if rank == 0:
+        offs = [0, 3, 5, 7, 8]
+        for r in range(4):
+            assert torch.equal(out[offs[r]:offs[r + 1]],
+                               full[offs[r]:offs[r + 1]] + r)
+    dist.destroy_process_group()
+
+
+def test_world4_gloo(tmp_path):
+    mp.spawn(_world4_worker, args=(4, str(tmp_path)), nprocs=4, join=True)
