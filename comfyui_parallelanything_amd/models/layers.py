@@ -85,18 +85,6 @@ class Modulation(nn.Module):
         return first, second
 
 
-def split_heads(x: torch.Tensor, num_heads: int) -> torch.Tensor:
-    """[B, S, H*D] -> [B, H, S, D]"""
-    B, S, _ = x.shape
-    return x.view(B, S, num_heads, -1).transpose(1, 2)
-
-
-def merge_heads(x: torch.Tensor) -> torch.Tensor:
-    """[B, H, S, D] -> [B, S, H*D]"""
-    B, H, S, D = x.shape
-    return x.transpose(1, 2).reshape(B, S, H * D)
-
-
 def rope_2d_table(
     h: int, w: int, axes_dim: Tuple[int, ...], theta: float = 10000.0,
     device=None, text_len: int = 0,

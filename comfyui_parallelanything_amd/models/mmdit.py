@@ -14,7 +14,7 @@ batch==1 layer-sharding works on these models unchanged.
 """
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Optional, Tuple
 
 import torch

@@ -11,7 +11,7 @@ engine intercept signature (reference any_device_parallel.py:1287).
 from __future__ import annotations
 
 import math
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import List, Optional, Tuple
 
 import torch

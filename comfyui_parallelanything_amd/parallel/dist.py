@@ -19,7 +19,7 @@ from __future__ import annotations
 
 import datetime
 import os
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Dict, List, Optional, Sequence
 
 import torch
