@@ -3,9 +3,8 @@
 // deltas are reliable despite cross-run/DVFS noise.
 //
 // Variants:
-//   0: shipped v4 (8-wave blocks, KVBLK 64)
-//   1: 128-key supertiles: stage 128 keys per barrier pair, run two
-//      sequential 64-key softmax passes (half the barrier/stage rounds)
+//   0: shipped v4 (8-wave blocks, KVBLK 64, 1-tile-deep prefetch)
+//   1: 2-tile-deep register prefetch (two kreg/vreg sets, ping-pong)
 //
 // Build: hipcc --offload-arch=gfx950 -O3 -std=c++17 attn_ab.hip -o attn_ab
 // Run:   ./attn_ab [rounds]
@@ -43,7 +42,8 @@ __global__ __launch_bounds__(512, 2) void attn_ab_kernel(
     int S, int Sk, float scale, int H) {
     constexpr int KVBLK = 64;
     constexpr int WAVES = 8;
-    constexpr int SUPER = (VAR == 1) ? 2 : 1;
+    constexpr int SUPER = 1;
+    constexpr int DEPTH = (VAR == 1) ? 2 : 1;
     constexpr int THREADS = WAVES * 64;
     constexpr int KPAD = D + 8;
     constexpr int VPAD = KVBLK + 8;
@@ -100,7 +100,7 @@ __global__ __launch_bounds__(512, 2) void attn_ab_kernel(
     float m_run = -1e30f, l_run = 0.f;
     const float scale2 = scale * PA_LOG2E;
 
-    bf16x8 kreg[KVECS], vreg[KVECS];
+    bf16x8 kreg[DEPTH][KVECS], vreg[DEPTH][KVECS];
     const bf16* kptr[KVECS];
     const bf16* vptr[KVECS];
     int lrow[KVECS];
@@ -116,35 +116,25 @@ __global__ __launch_bounds__(512, 2) void attn_ab_kernel(
         }
     }
 
-    auto issue_tile_loads = [&](int kv0) {
+    auto issue_tile_loads = [&](int kv0, int slot) {
 #pragma unroll
         for (int i = 0; i < KVECS; ++i) {
-            if (false) {
-                if (kv0 + lrow[i] < Sk) {
-                    kreg[i] = *reinterpret_cast<const bf16x8*>(kptr[i]);
-                    vreg[i] = *reinterpret_cast<const bf16x8*>(vptr[i]);
-                } else {
-                    kreg[i] = bf16x8{0,0,0,0,0,0,0,0};
-                    vreg[i] = bf16x8{0,0,0,0,0,0,0,0};
-                }
-                kptr[i] += (long)KVBLK * ss;
-                vptr[i] += (long)KVBLK * ss;
-            } else {
+            {
                 const int idx = tid + i * THREADS;
                 const int row = idx / (D / 8);
                 const int col = (idx % (D / 8)) * 8;
                 const int src = kv0 + row;
                 if (src < Sk) {
-                    kreg[i] = *reinterpret_cast<const bf16x8*>(kp + (long)src * ss + col);
-                    vreg[i] = *reinterpret_cast<const bf16x8*>(vp + (long)src * ss + col);
+                    kreg[slot][i] = *reinterpret_cast<const bf16x8*>(kp + (long)src * ss + col);
+                    vreg[slot][i] = *reinterpret_cast<const bf16x8*>(vp + (long)src * ss + col);
                 } else {
-                    kreg[i] = bf16x8{0,0,0,0,0,0,0,0};
-                    vreg[i] = bf16x8{0,0,0,0,0,0,0,0};
+                    kreg[slot][i] = bf16x8{0,0,0,0,0,0,0,0};
+                    vreg[slot][i] = bf16x8{0,0,0,0,0,0,0,0};
                 }
             }
         }
     };
-    auto write_tile_lds = [&]() {
+    auto write_tile_lds = [&](int slot) {
 #pragma unroll
         for (int i = 0; i < KVECS; ++i) {
             const int idx = tid + i * THREADS;
@@ -152,7 +142,7 @@ __global__ __launch_bounds__(512, 2) void attn_ab_kernel(
             const int col = (idx % (D / 8)) * 8;
             const int sub = row / KVBLK;          // 64-key sub-tile id
             const int srow = row % KVBLK;
-            *reinterpret_cast<bf16x8*>(&k_lds[row * KPAD + col]) = kreg[i];
+            *reinterpret_cast<bf16x8*>(&k_lds[row * KPAD + col]) = kreg[slot][i];
             if (VPAIR && D == 128) {
                 // pair keys (row, row^1) via shfl_xor(16): even-key threads
                 // write dims col..col+3 as b32, odd-key threads col+4..col+7
@@ -161,9 +151,9 @@ __global__ __launch_bounds__(512, 2) void attn_ab_kernel(
                 for (int jj = 0; jj < 4; ++jj) {
                     const int j = even ? jj : jj + 4;
                     const unsigned short mine =
-                        (unsigned short)vreg[i][j];
+                        (unsigned short)vreg[slot][i][j];
                     const unsigned short partner = (unsigned short)__shfl_xor(
-                        (int)(unsigned short)vreg[i][j ^ 4], 16, 64);
+                        (int)(unsigned short)vreg[slot][i][j ^ 4], 16, 64);
                     const unsigned short lo_key = even ? mine : partner;
                     const unsigned short hi_key = even ? partner : mine;
                     const int dim = col + j;
@@ -181,7 +171,7 @@ __global__ __launch_bounds__(512, 2) void attn_ab_kernel(
                     const int key_swz =
                         (((srow >> 3) ^ ((dim >> 3) & 7)) << 3) | (srow & 7);
                     v_lds[(sub * D + dim) * VPAD + key_swz] =
-                        __ushort_as_bfloat16((unsigned short)vreg[i][j]);
+                        __ushort_as_bfloat16((unsigned short)vreg[slot][i][j]);
                 }
             }
         }
@@ -287,17 +277,16 @@ __global__ __launch_bounds__(512, 2) void attn_ab_kernel(
         if (!NO_PRIO) __builtin_amdgcn_s_setprio(0);
     };
 
-    const int SBLK = SUPER * KVBLK;
-    const int n_super = (Sk + SBLK - 1) / SBLK;
-    issue_tile_loads(0);
-    for (int t = 0; t < n_super; ++t) {
+    const int n_tiles = (Sk + KVBLK - 1) / KVBLK;
+    issue_tile_loads(0, 0);
+    if (DEPTH == 2 && n_tiles > 1) issue_tile_loads(KVBLK, 1);
+    for (int t = 0; t < n_tiles; ++t) {
         __syncthreads();
-        write_tile_lds();
+        write_tile_lds(t % DEPTH);
         __syncthreads();
-        if (t + 1 < n_super) issue_tile_loads((t + 1) * SBLK);
-#pragma unroll
-        for (int sub = 0; sub < SUPER; ++sub)
-            tile(t * SBLK + sub * KVBLK, true, sub);
+        if (t + DEPTH < n_tiles || (DEPTH == 1 && t + 1 < n_tiles))
+            issue_tile_loads((t + DEPTH) * KVBLK, (t + DEPTH) % DEPTH);
+        tile(t * KVBLK, true, 0);
     }
 
     const int row = q0 + l32;
