@@ -56,6 +56,9 @@ def parse_args():
                     help="tiny config (CPU debugging only)")
     ap.add_argument("--no-graph", dest="graph", action="store_false",
                     help="disable hipGraph capture of the per-rank forward")
+    ap.add_argument("--weights", default=None,
+                    help="comma-separated per-rank split weights (e.g. 60,40 "
+                         "for the Z-Image headline config); default even")
     ap.add_argument("--json-out", default=None)
     return ap.parse_args()
 
@@ -88,7 +91,14 @@ def main():
 
         quantize_fp8(model)
 
-    sizes = compute_split_sizes(args.batch, [1.0 / n] * n)
+    if args.weights:
+        ws = [float(w) for w in args.weights.split(",")]
+        if len(ws) != n:
+            raise SystemExit("--weights count must equal the rank count")
+        total_w = sum(ws)
+        sizes = compute_split_sizes(args.batch, [w / total_w for w in ws])
+    else:
+        sizes = compute_split_sizes(args.batch, [1.0 / n] * n)
     if args.model == "wan":
         x, t, ctx, kw = make_inputs(args.batch, dev=dev, dtype=dtype, tiny=tiny)
     else:
@@ -188,7 +198,8 @@ def main():
                 "global_batch": args.batch,
                 "resolution": args.px,
                 "seq_len": (args.px // 16) ** 2 + 512 if args.model == "flux" else None,
-                "parallelism": f"dp{n}",
+                "parallelism": f"dp{n}" + (
+                    f" weighted {args.weights}" if args.weights else ""),
             },
         }
         line = json.dumps(result)
