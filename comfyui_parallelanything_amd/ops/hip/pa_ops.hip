@@ -990,61 +990,72 @@ __global__ void pack_joint_qkv_kernel(
     long t_bs, long t_ss, long t_qs, long t_hs,     // txt strides (b, s, qkv, h)
     long i_bs, long i_ss, long i_qs, long i_hs,
     long n_rows, float eps) {
+    // RPW rows per wave, all 3*RPW loads issued before the reductions
+    // (single-row version measured latency-bound like qk_norm_rope).
+    constexpr int RPW = 4;
     const int lane = threadIdx.x & 63;
-    const long row = ((long)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
-    if (row >= n_rows) return;
+    const long wave = ((long)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
     const int S = T + Si;
-    const long b = row / ((long)S * H);
-    const long sh = row % ((long)S * H);
-    const int sj = (int)(sh / H);
-    const int h = (int)(sh % H);
     const int pairs = D / 2;
     if (lane >= pairs) return;
 
-    const bool is_txt = sj < T;
-    const bf16* src = is_txt
-        ? txt + b * t_bs + (long)sj * t_ss + (long)h * t_hs
-        : img + b * i_bs + (long)(sj - T) * i_ss + (long)h * i_hs;
-    const long qs = is_txt ? t_qs : i_qs;
-    const unsigned int uwq = reinterpret_cast<const unsigned int*>(
-        is_txt ? wq_t : wq_i)[lane];
-    const unsigned int uwk = reinterpret_cast<const unsigned int*>(
-        is_txt ? wk_t : wk_i)[lane];
-    const float c = cs[((long)sj * pairs + lane) * 2 + 0];
-    const float sn = cs[((long)sj * pairs + lane) * 2 + 1];
-
-    const unsigned int uq = reinterpret_cast<const unsigned int*>(src)[lane];
-    const unsigned int uk = reinterpret_cast<const unsigned int*>(src + qs)[lane];
-    const unsigned int uv = reinterpret_cast<const unsigned int*>(src + 2 * qs)[lane];
-
-    const long obase = (((long)b * S + sj) * H + h) * (D / 2);
-    {   // q: rms + weight + rope
-        float a0 = bf2f(__ushort_as_bfloat16((unsigned short)(uq & 0xffff)));
-        float a1 = bf2f(__ushort_as_bfloat16((unsigned short)(uq >> 16)));
-        float ss_ = a0 * a0 + a1 * a1;
+    unsigned int uq[RPW], uk[RPW], uv[RPW], uwq[RPW], uwk[RPW];
+    long obase[RPW];
+    float cvec[RPW], svec[RPW];
 #pragma unroll
-        for (int off = 32; off > 0; off >>= 1) ss_ += __shfl_xor(ss_, off, 64);
-        const float rr = rsqrtf(ss_ / (float)D + eps);
-        a0 = a0 * rr * bf2f(__ushort_as_bfloat16((unsigned short)(uwq & 0xffff)));
-        a1 = a1 * rr * bf2f(__ushort_as_bfloat16((unsigned short)(uwq >> 16)));
-        reinterpret_cast<unsigned int*>(oq)[obase + lane] =
-            (unsigned int)__bfloat16_as_ushort(f2bf(a0 * c - a1 * sn)) |
-            ((unsigned int)__bfloat16_as_ushort(f2bf(a0 * sn + a1 * c)) << 16);
+    for (int i = 0; i < RPW; ++i) {
+        const long row = wave * RPW + i;
+        const long rr = row < n_rows ? row : n_rows - 1;
+        const long b = rr / ((long)S * H);
+        const long sh = rr % ((long)S * H);
+        const int sj = (int)(sh / H);
+        const int h = (int)(sh % H);
+        const bool is_txt = sj < T;
+        const bf16* src = is_txt
+            ? txt + b * t_bs + (long)sj * t_ss + (long)h * t_hs
+            : img + b * i_bs + (long)(sj - T) * i_ss + (long)h * i_hs;
+        const long qs = is_txt ? t_qs : i_qs;
+        uq[i] = reinterpret_cast<const unsigned int*>(src)[lane];
+        uk[i] = reinterpret_cast<const unsigned int*>(src + qs)[lane];
+        uv[i] = reinterpret_cast<const unsigned int*>(src + 2 * qs)[lane];
+        uwq[i] = reinterpret_cast<const unsigned int*>(is_txt ? wq_t : wq_i)[lane];
+        uwk[i] = reinterpret_cast<const unsigned int*>(is_txt ? wk_t : wk_i)[lane];
+        cvec[i] = cs[((long)sj * pairs + lane) * 2 + 0];
+        svec[i] = cs[((long)sj * pairs + lane) * 2 + 1];
+        obase[i] = (((long)b * S + sj) * H + h) * (D / 2);
     }
-    {   // k
-        float a0 = bf2f(__ushort_as_bfloat16((unsigned short)(uk & 0xffff)));
-        float a1 = bf2f(__ushort_as_bfloat16((unsigned short)(uk >> 16)));
-        float ss_ = a0 * a0 + a1 * a1;
 #pragma unroll
-        for (int off = 32; off > 0; off >>= 1) ss_ += __shfl_xor(ss_, off, 64);
-        const float rr = rsqrtf(ss_ / (float)D + eps);
-        a0 = a0 * rr * bf2f(__ushort_as_bfloat16((unsigned short)(uwk & 0xffff)));
-        a1 = a1 * rr * bf2f(__ushort_as_bfloat16((unsigned short)(uwk >> 16)));
-        reinterpret_cast<unsigned int*>(ok)[obase + lane] =
-            (unsigned int)__bfloat16_as_ushort(f2bf(a0 * c - a1 * sn)) |
-            ((unsigned int)__bfloat16_as_ushort(f2bf(a0 * sn + a1 * c)) << 16);
+    for (int i = 0; i < RPW; ++i) {
+        if (wave * RPW + i >= n_rows) break;
+        const float c = cvec[i], sn = svec[i];
+        {   // q: rms + weight + rope
+            float a0 = bf2f(__ushort_as_bfloat16((unsigned short)(uq[i] & 0xffff)));
+            float a1 = bf2f(__ushort_as_bfloat16((unsigned short)(uq[i] >> 16)));
+            float ss_ = a0 * a0 + a1 * a1;
+#pragma unroll
+            for (int off = 32; off > 0; off >>= 1) ss_ += __shfl_xor(ss_, off, 64);
+            const float rr = rsqrtf(ss_ / (float)D + eps);
+            a0 = a0 * rr * bf2f(__ushort_as_bfloat16((unsigned short)(uwq[i] & 0xffff)));
+            a1 = a1 * rr * bf2f(__ushort_as_bfloat16((unsigned short)(uwq[i] >> 16)));
+            reinterpret_cast<unsigned int*>(oq)[obase[i] + lane] =
+                (unsigned int)__bfloat16_as_ushort(f2bf(a0 * c - a1 * sn)) |
+                ((unsigned int)__bfloat16_as_ushort(f2bf(a0 * sn + a1 * c)) << 16);
+        }
+        {   // k
+            float a0 = bf2f(__ushort_as_bfloat16((unsigned short)(uk[i] & 0xffff)));
+            float a1 = bf2f(__ushort_as_bfloat16((unsigned short)(uk[i] >> 16)));
+            float ss_ = a0 * a0 + a1 * a1;
+#pragma unroll
+            for (int off = 32; off > 0; off >>= 1) ss_ += __shfl_xor(ss_, off, 64);
+            const float rr = rsqrtf(ss_ / (float)D + eps);
+            a0 = a0 * rr * bf2f(__ushort_as_bfloat16((unsigned short)(uwk[i] & 0xffff)));
+            a1 = a1 * rr * bf2f(__ushort_as_bfloat16((unsigned short)(uwk[i] >> 16)));
+            reinterpret_cast<unsigned int*>(ok)[obase[i] + lane] =
+                (unsigned int)__bfloat16_as_ushort(f2bf(a0 * c - a1 * sn)) |
+                ((unsigned int)__bfloat16_as_ushort(f2bf(a0 * sn + a1 * c)) << 16);
+        }
+        reinterpret_cast<unsigned int*>(ov)[obase[i] + lane] = uv[i];
     }
-    reinterpret_cast<unsigned int*>(ov)[obase + lane] = uv;  // v passthrough
 }
 
 // ---------------------------------------------------------------------------
@@ -1367,7 +1378,7 @@ std::vector<at::Tensor> pack_joint_qkv(at::Tensor txt_qkv, at::Tensor img_qkv,
     auto ok = at::empty({B, T + Si, H, D}, opts);
     auto ov = at::empty({B, T + Si, H, D}, opts);
     const long rows = (long)B * (T + Si) * H;
-    const long blocks = (rows * 64 + 255) / 256;
+    const long blocks = (((rows + 3) / 4) * 64 + 255) / 256;
     hipLaunchKernelGGL(pack_joint_qkv_kernel, dim3((unsigned)blocks), dim3(256),
                        0, cur_stream(),
                        (const bf16*)txt_qkv.data_ptr(),
