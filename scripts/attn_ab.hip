@@ -351,8 +351,27 @@ int main(int argc, char** argv) {
                 hipLaunchKernelGGL((attn_ab_kernel<128, 1>), grid, blk, 0, 0,
                                    q, k, v, o, S, S, scale, H);
         };
-        for (int var = 0; var < 2; ++var) { run(var); HIP_CHECK(hipGetLastError()); }
+        // correctness gate: variants must agree with variant 0
+        std::vector<unsigned short> ref(4096), got(4096);
+        run(0); HIP_CHECK(hipGetLastError());
         HIP_CHECK(hipDeviceSynchronize());
+        HIP_CHECK(hipMemcpy(ref.data(), o, 4096 * 2, hipMemcpyDeviceToHost));
+        for (int var = 1; var < 2; ++var) {
+            HIP_CHECK(hipMemset(o, 0, n * 2));
+            run(var); HIP_CHECK(hipGetLastError());
+            HIP_CHECK(hipDeviceSynchronize());
+            HIP_CHECK(hipMemcpy(got.data(), o, 4096 * 2, hipMemcpyDeviceToHost));
+            int bad = 0;
+            for (int i = 0; i < 4096; ++i) {
+                float a, b2;
+                unsigned int ua = (unsigned int)ref[i] << 16,
+                             ub = (unsigned int)got[i] << 16;
+                __builtin_memcpy(&a, &ua, 4);
+                __builtin_memcpy(&b2, &ub, 4);
+                if (fabsf(a - b2) > 0.05f + 0.05f * fabsf(a)) ++bad;
+            }
+            if (bad) printf("VARIANT %d WRONG: %d/4096 mismatches\n", var, bad);
+        }
         double best[2] = {1e30, 1e30};
         for (int r = 0; r < rounds; ++r)
             for (int var = 0; var < 2; ++var) {

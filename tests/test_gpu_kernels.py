@@ -274,3 +274,27 @@ def test_gelu_tanh_strided_slice():
     out = ops.gelu_tanh(mlp)
     ref = torch.nn.functional.gelu(mlp.float(), approximate="tanh")
     _cmp(out, ref, 2e-2, 2e-2, "gelu strided")
+
+
+def test_attn_v3_fallback_env():
+    """PA_ATTN_V3=1 selects the previous-generation kernel; numerics hold."""
+    import os
+    import subprocess
+    import sys
+
+    code = (
+        "import torch;"
+        "from comfyui_parallelanything_amd import ops;"
+        "from comfyui_parallelanything_amd.ops import reference as R;"
+        "torch.manual_seed(0);"
+        "q=torch.randn(1,2,256,128,device='cuda',dtype=torch.bfloat16);"
+        "k=torch.randn_like(q); v=torch.randn_like(q);"
+        "out=ops.attention(q,k,v);"
+        "ref=R.attention(q.float(),k.float(),v.float());"
+        "import sys;"
+        "sys.exit(0 if (out.float()-ref).abs().max().item()<0.05 else 1)"
+    )
+    env = dict(os.environ, PA_ATTN_V3="1")
+    r = subprocess.run([sys.executable, "-c", code], env=env,
+                       cwd=os.path.dirname(os.path.dirname(__file__)))
+    assert r.returncode == 0
