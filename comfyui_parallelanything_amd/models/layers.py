@@ -163,9 +163,12 @@ class DoubleStreamBlock(nn.Module):
         )
         return q, k, v
 
-    def forward(self, img, txt, vec, pe):
-        img_m1, img_m2 = self.img_mod(vec)
-        txt_m1, txt_m2 = self.txt_mod(vec)
+    def forward(self, img, txt, vec, pe, mods=None):
+        if mods is not None:
+            (img_m1, img_m2), (txt_m1, txt_m2) = mods
+        else:
+            img_m1, img_m2 = self.img_mod(vec)
+            txt_m1, txt_m2 = self.txt_mod(vec)
 
         img_in = ops.layer_norm_mod(img, img_m1.scale, img_m1.shift)
         txt_in = ops.layer_norm_mod(txt, txt_m1.scale, txt_m1.shift)
@@ -217,9 +220,12 @@ class SingleStreamBlock(nn.Module):
         self.attn = JointAttention(num_heads, head_dim)
         self.mlp_act = FusedGELU()
 
-    def forward(self, x, vec, pe):
+    def forward(self, x, vec, pe, mods=None):
         B, S, hidden = x.shape
-        mod, _ = self.modulation(vec)
+        if mods is not None:
+            (mod,) = mods
+        else:
+            mod, _ = self.modulation(vec)
         x_in = ops.layer_norm_mod(x, mod.scale, mod.shift)
         proj = self.linear1(x_in)
         qkv = proj[..., : 3 * hidden].unflatten(-1, (3, self.num_heads, -1))
@@ -244,3 +250,47 @@ class LastLayer(nn.Module):
     def forward(self, x, vec):
         shift, scale = self.ada_lin(torch.nn.functional.silu(vec)).chunk(2, dim=-1)
         return self.linear(ops.layer_norm_mod(x, scale, shift))
+
+
+def _mod_outs(chunked):
+    """list of 3-chunk groups -> ModOut tuples."""
+    return ModOut(*chunked[:3]), (ModOut(*chunked[3:]) if len(chunked) == 6 else None)
+
+
+class ModulationBank:
+    """One fused GEMM for every per-block AdaLN projection.
+
+    All Modulation.lin layers (and nothing else) consume silu(vec), so their
+    weights concatenate into a single [sum_out, hidden] matrix: one
+    hipBLASLt call per step replaces ~115 tiny GEMMs + silu launches. The
+    concatenated weights are a per-replica cache (CACHE_ATTRS clears it on
+    replication; it rebuilds lazily from the replica's own - possibly
+    LoRA-patched - weights on its own device).
+    """
+
+    def __init__(self, mod_modules):
+        self.mods = list(mod_modules)  # Modulation instances, block order
+        self.weight = None
+        self.bias = None
+        self.splits = [m.multiplier * m.lin.in_features // 1 for m in self.mods]
+        self.splits = [m.lin.out_features for m in self.mods]
+
+    def _build(self):
+        self.weight = torch.cat([m.lin.weight for m in self.mods], dim=0)
+        self.bias = torch.cat([m.lin.bias for m in self.mods], dim=0)
+
+    def __call__(self, vec):
+        if self.weight is None or self.weight.device != vec.device:
+            self._build()
+        out = torch.nn.functional.linear(
+            torch.nn.functional.silu(vec), self.weight, self.bias
+        )
+        results = []
+        off = 0
+        for m, width in zip(self.mods, self.splits):
+            seg = out[:, off:off + width]
+            off += width
+            dim = m.lin.in_features
+            parts = [seg[:, i * dim:(i + 1) * dim] for i in range(m.multiplier)]
+            results.append(_mod_outs(parts))
+        return results

@@ -25,9 +25,16 @@ from .layers import (
     DoubleStreamBlock,
     LastLayer,
     MLPEmbedder,
+    ModulationBank,
     SingleStreamBlock,
     rope_2d_table,
 )
+
+
+def _base_block(blk):
+    """Reach through pipeline wrappers (ParallelBlock.local_block /
+    DistPipelineBlock.block) to the underlying block."""
+    return getattr(blk, "local_block", None) or getattr(blk, "block", None) or blk
 
 
 @dataclass
@@ -86,6 +93,7 @@ class Flux(nn.Module):
         # per-replica RoPE cache, keyed by (h, w, txt_len); lives on the
         # owning device by construction (vs reference clear_flux_caches).
         self._pe_cache: dict = {}
+        self._mod_cache: dict = {}  # ModulationBank per replica (CACHE_ATTRS)
 
     def _patchify(self, x: torch.Tensor):
         B, C, H, W = x.shape
@@ -139,11 +147,26 @@ class Flux(nn.Module):
             )
         pe = self._pe(h, w, txt.shape[1], x.device, x.dtype)
 
+        bank = self._mod_cache.get("bank")
+        if bank is None:
+            mods = []
+            for blk in self.double_blocks:
+                b = _base_block(blk)
+                mods += [b.img_mod, b.txt_mod]
+            mods += [_base_block(blk).modulation for blk in self.single_blocks]
+            bank = ModulationBank(mods)
+            self._mod_cache["bank"] = bank
+        banked = bank(vec)
+
+        i = 0
         for block in self.double_blocks:
-            img, txt = block(img, txt, vec, pe)
+            img, txt = block(img, txt, vec, pe,
+                             mods=(banked[i], banked[i + 1]))
+            i += 2
         xcat = torch.cat([txt, img], dim=1)
         for block in self.single_blocks:
-            xcat = block(xcat, vec, pe)
+            xcat = block(xcat, vec, pe, mods=(banked[i][0],))
+            i += 1
         img = xcat[:, txt.shape[1]:]
         out = self.final_layer(img, vec)
         return self._unpatchify(out, h, w)
@@ -193,6 +216,7 @@ class ZImage(nn.Module):
         )
         self.final_layer = LastLayer(cfg.hidden, self.patch_dim)
         self._pe_cache: dict = {}
+        self._mod_cache: dict = {}
 
     def _pe(self, h, w, txt_len, device):
         key = (h, w, txt_len, str(device))
@@ -220,8 +244,15 @@ class ZImage(nn.Module):
         )
         seq = torch.cat([txt, img], dim=1)
         pe = self._pe(h, w, txt.shape[1], x.device)
-        for block in self.layers:
-            seq = block(seq, vec, pe)
+        bank = self._mod_cache.get("bank")
+        if bank is None:
+            bank = ModulationBank(
+                [_base_block(blk).modulation for blk in self.layers]
+            )
+            self._mod_cache["bank"] = bank
+        banked = bank(vec)
+        for i, block in enumerate(self.layers):
+            seq = block(seq, vec, pe, mods=(banked[i][0],))
         out = self.final_layer(seq[:, txt.shape[1]:], vec)
         return (
             out.view(B, h, w, C, p, p)

@@ -43,7 +43,7 @@ def _iter_tensors(module: nn.Module):
 # any_device_parallel.py:166-195); our models keep per-device caches in
 # dicts with these names, which are simply reset on the new replica so it
 # repopulates on its own GPU.
-CACHE_ATTRS = ("_pe_cache",)
+CACHE_ATTRS = ("_pe_cache", "_mod_cache")
 
 
 def clear_replica_caches(module: nn.Module) -> int:
