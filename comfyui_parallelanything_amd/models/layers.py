@@ -270,9 +270,12 @@ class ModulationBank:
 
     def __init__(self, mod_modules):
         self.mods = list(mod_modules)  # Modulation instances, block order
+        if not all(isinstance(m.lin, nn.Linear) for m in self.mods):
+            # quantized/wrapped projections: banking disabled, blocks
+            # compute their own modulation (mods=None path)
+            raise TypeError("ModulationBank requires plain nn.Linear mods")
         self.weight = None
         self.bias = None
-        self.splits = [m.multiplier * m.lin.in_features // 1 for m in self.mods]
         self.splits = [m.lin.out_features for m in self.mods]
 
     def _build(self):

@@ -149,23 +149,31 @@ class Flux(nn.Module):
 
         bank = self._mod_cache.get("bank")
         if bank is None:
-            mods = []
-            for blk in self.double_blocks:
-                b = _base_block(blk)
-                mods += [b.img_mod, b.txt_mod]
-            mods += [_base_block(blk).modulation for blk in self.single_blocks]
-            bank = ModulationBank(mods)
+            try:
+                mods = []
+                for blk in self.double_blocks:
+                    b = _base_block(blk)
+                    mods += [b.img_mod, b.txt_mod]
+                mods += [
+                    _base_block(blk).modulation for blk in self.single_blocks
+                ]
+                bank = ModulationBank(mods)
+            except TypeError:
+                bank = False  # quantized mods: per-block path
             self._mod_cache["bank"] = bank
-        banked = bank(vec)
+        banked = bank(vec) if bank else None
 
         i = 0
         for block in self.double_blocks:
-            img, txt = block(img, txt, vec, pe,
-                             mods=(banked[i], banked[i + 1]))
+            img, txt = block(
+                img, txt, vec, pe,
+                mods=(banked[i], banked[i + 1]) if banked else None,
+            )
             i += 2
         xcat = torch.cat([txt, img], dim=1)
         for block in self.single_blocks:
-            xcat = block(xcat, vec, pe, mods=(banked[i][0],))
+            xcat = block(xcat, vec, pe,
+                         mods=(banked[i][0],) if banked else None)
             i += 1
         img = xcat[:, txt.shape[1]:]
         out = self.final_layer(img, vec)
@@ -246,13 +254,17 @@ class ZImage(nn.Module):
         pe = self._pe(h, w, txt.shape[1], x.device)
         bank = self._mod_cache.get("bank")
         if bank is None:
-            bank = ModulationBank(
-                [_base_block(blk).modulation for blk in self.layers]
-            )
+            try:
+                bank = ModulationBank(
+                    [_base_block(blk).modulation for blk in self.layers]
+                )
+            except TypeError:
+                bank = False
             self._mod_cache["bank"] = bank
-        banked = bank(vec)
+        banked = bank(vec) if bank else None
         for i, block in enumerate(self.layers):
-            seq = block(seq, vec, pe, mods=(banked[i][0],))
+            seq = block(seq, vec, pe,
+                        mods=(banked[i][0],) if banked else None)
         out = self.final_layer(seq[:, txt.shape[1]:], vec)
         return (
             out.view(B, h, w, C, p, p)

@@ -50,6 +50,8 @@ class FP8Linear(nn.Module):
         else:
             self.bias = None
         self.out_dtype = out_dtype
+        self.in_features = weight_fp8.shape[1]
+        self.out_features = weight_fp8.shape[0]
 
     @classmethod
     def from_linear(cls, lin: nn.Linear) -> "FP8Linear":
