@@ -29,6 +29,36 @@ class FusedGELU(nn.Module):
         return ops.gelu_tanh(x)
 
 
+class GELULinear(nn.Module):
+    """Linear with the GELU fused into the GEMM epilogue (hipBLASLt
+    _addmm_activation on GPU; linear+tanh-gelu on CPU)."""
+
+    def __init__(self, in_features: int, out_features: int):
+        super().__init__()
+        self.lin = nn.Linear(in_features, out_features, bias=True)
+
+    def forward(self, x):
+        if x.is_cuda:
+            x2 = x.reshape(-1, x.shape[-1])
+            out = torch._addmm_activation(
+                self.lin.bias, x2, self.lin.weight.t(), use_gelu=True
+            )
+            return out.reshape(*x.shape[:-1], out.shape[-1])
+        return torch.nn.functional.gelu(self.lin(x), approximate="tanh")
+
+
+class FusedMLP(nn.Module):
+    """Linear -> GELU (epilogue-fused) -> Linear."""
+
+    def __init__(self, dim_in: int, dim_mid: int, dim_out: int):
+        super().__init__()
+        self.up = GELULinear(dim_in, dim_mid)
+        self.down = nn.Linear(dim_mid, dim_out, bias=True)
+
+    def forward(self, x):
+        return self.down(self.up(x))
+
+
 class MLPEmbedder(nn.Module):
     """2-layer SiLU MLP used for timestep / vector conditioning."""
 
@@ -138,18 +168,12 @@ class DoubleStreamBlock(nn.Module):
         self.img_attn_qkv = nn.Linear(hidden, hidden * 3)
         self.img_attn_norm = QKNorm(head_dim)
         self.img_attn_proj = nn.Linear(hidden, hidden)
-        self.img_mlp = nn.Sequential(
-            nn.Linear(hidden, mlp_dim), FusedGELU(),
-            nn.Linear(mlp_dim, hidden),
-        )
+        self.img_mlp = FusedMLP(hidden, mlp_dim, hidden)
         self.txt_mod = Modulation(hidden, double=True)
         self.txt_attn_qkv = nn.Linear(hidden, hidden * 3)
         self.txt_attn_norm = QKNorm(head_dim)
         self.txt_attn_proj = nn.Linear(hidden, hidden)
-        self.txt_mlp = nn.Sequential(
-            nn.Linear(hidden, mlp_dim), FusedGELU(),
-            nn.Linear(mlp_dim, hidden),
-        )
+        self.txt_mlp = FusedMLP(hidden, mlp_dim, hidden)
         self.attn = JointAttention(num_heads, head_dim)
 
     def _qkv(self, x, qkv_layer, norm, pe_slice):
@@ -210,15 +234,16 @@ class SingleStreamBlock(nn.Module):
         head_dim = hidden // num_heads
         self.num_heads = num_heads
         self.mlp_dim = int(hidden * mlp_ratio)
-        self.linear1 = nn.Linear(hidden, hidden * 3 + self.mlp_dim)
-        # split output projection: linear2([attn | gelu(mlp)]) as two GEMMs —
-        # avoids materializing a strided last-dim concat every block
+        # split projections: qkv separately from the MLP up-projection so
+        # the GELU fuses into the MLP GEMM epilogue; the output projection is
+        # split the same way (no strided last-dim concats anywhere)
+        self.linear1_qkv = nn.Linear(hidden, hidden * 3)
+        self.linear1_mlp = GELULinear(hidden, self.mlp_dim)
         self.linear2_attn = nn.Linear(hidden, hidden)
         self.linear2_mlp = nn.Linear(self.mlp_dim, hidden, bias=False)
         self.norm = QKNorm(head_dim)
         self.modulation = Modulation(hidden, double=False)
         self.attn = JointAttention(num_heads, head_dim)
-        self.mlp_act = FusedGELU()
 
     def forward(self, x, vec, pe, mods=None):
         B, S, hidden = x.shape
@@ -227,15 +252,14 @@ class SingleStreamBlock(nn.Module):
         else:
             mod, _ = self.modulation(vec)
         x_in = ops.layer_norm_mod(x, mod.scale, mod.shift)
-        proj = self.linear1(x_in)
-        qkv = proj[..., : 3 * hidden].unflatten(-1, (3, self.num_heads, -1))
-        mlp_in = proj[..., 3 * hidden:]
-        q, k, v = qkv.unbind(2)  # [B,S,H,D] views into proj
+        qkv = self.linear1_qkv(x_in).unflatten(-1, (3, self.num_heads, -1))
+        mlp_act = self.linear1_mlp(x_in)  # GELU in the GEMM epilogue
+        q, k, v = qkv.unbind(2)  # [B,S,H,D] views
         ops.qk_norm_rope_(
             q, k, self.norm.query_norm.scale, self.norm.key_norm.scale, pe
         )
         attn = ops.attention_bshd(q, k, v, self.attn.scale).flatten(2)
-        out = self.linear2_attn(attn) + self.linear2_mlp(self.mlp_act(mlp_in))
+        out = self.linear2_attn(attn) + self.linear2_mlp(mlp_act)
         return ops.gate_residual(x, mod.gate, out)
 
 

@@ -16,7 +16,7 @@ import torch
 from torch import nn
 
 from .. import ops
-from .layers import FusedGELU, MLPEmbedder, QKNorm
+from .layers import FusedMLP, MLPEmbedder, QKNorm
 
 
 def rope_3d_table(f: int, h: int, w: int, axes_dim: Tuple[int, ...],
@@ -47,10 +47,7 @@ class WanBlock(nn.Module):
         self.cross_k = nn.Linear(ctx_dim, dim)
         self.cross_v = nn.Linear(ctx_dim, dim)
         self.cross_proj = nn.Linear(dim, dim)
-        self.ffn = nn.Sequential(
-            nn.Linear(dim, ffn_dim), FusedGELU(),
-            nn.Linear(ffn_dim, dim),
-        )
+        self.ffn = FusedMLP(dim, ffn_dim, dim)
 
     def forward(self, x, e, context, pe):
         # e: [B, 6, dim] time-modulation; learned bias added per block
@@ -118,10 +115,7 @@ class WanDiT(nn.Module):
         pf, ph, pw = cfg.patch_size
         self.patch_dim = cfg.in_channels * pf * ph * pw
         self.patch_in = nn.Linear(self.patch_dim, cfg.dim)
-        self.txt_in = nn.Sequential(
-            nn.Linear(cfg.ctx_dim, cfg.dim), FusedGELU(),
-            nn.Linear(cfg.dim, cfg.dim),
-        )
+        self.txt_in = FusedMLP(cfg.ctx_dim, cfg.dim, cfg.dim)
         self.time_in = MLPEmbedder(cfg.time_embed_dim, cfg.dim)
         self.time_proj = nn.Linear(cfg.dim, cfg.dim * 6)
         self.transformer_blocks = nn.ModuleList(
