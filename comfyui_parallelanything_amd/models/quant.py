@@ -89,6 +89,10 @@ def quantize_fp8(model: nn.Module, min_features: int = 1024) -> int:
         )
     n = 0
     for parent in model.modules():
+        if type(parent).__name__ == "GELULinear":
+            # keep the GELU-epilogue GEMM in bf16: _scaled_mm has no gelu
+            # epilogue, so quantizing would un-fuse the activation
+            continue
         for name, child in list(parent.named_children()):
             if isinstance(child, nn.Linear) and (
                 child.in_features >= min_features
