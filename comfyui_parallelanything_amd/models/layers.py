@@ -259,7 +259,16 @@ class SingleStreamBlock(nn.Module):
             q, k, self.norm.query_norm.scale, self.norm.key_norm.scale, pe
         )
         attn = ops.attention_bshd(q, k, v, self.attn.scale).flatten(2)
-        out = self.linear2_attn(attn) + self.linear2_mlp(mlp_act)
+        if x.is_cuda:
+            # second GEMM accumulates into the first's output (beta=1
+            # epilogue) — no separate elementwise add
+            acc = self.linear2_attn(attn).reshape(-1, hidden)
+            out = torch.addmm(
+                acc, mlp_act.reshape(-1, self.mlp_dim),
+                self.linear2_mlp.weight.t(),
+            ).reshape(B, S, hidden)
+        else:
+            out = self.linear2_attn(attn) + self.linear2_mlp(mlp_act)
         return ops.gate_residual(x, mod.gate, out)
 
 
