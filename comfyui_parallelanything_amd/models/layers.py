@@ -259,7 +259,7 @@ class SingleStreamBlock(nn.Module):
             q, k, self.norm.query_norm.scale, self.norm.key_norm.scale, pe
         )
         attn = ops.attention_bshd(q, k, v, self.attn.scale).flatten(2)
-        if x.is_cuda:
+        if x.is_cuda and isinstance(self.linear2_mlp, nn.Linear):
             # second GEMM accumulates into the first's output (beta=1
             # epilogue) — no separate elementwise add
             acc = self.linear2_attn(attn).reshape(-1, hidden)
