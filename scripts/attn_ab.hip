@@ -3,8 +3,10 @@
 // deltas are reliable despite cross-run/DVFS noise.
 //
 // Variants:
-//   0: shipped v4 (8-wave blocks, KVBLK 64, 1-tile-deep prefetch)
-//   1: 2-tile-deep register prefetch (two kreg/vreg sets, ping-pong)
+//   0: shipped v4
+//   1: + nt epilogue O stores + nt Q loads (keep K/V resident in the
+//        XCD's L2; O is written once, Q read once per block)
+//      + static s_setprio(1) for the younger wave half (guide T5 static)
 //
 // Build: hipcc --offload-arch=gfx950 -O3 -std=c++17 attn_ab.hip -o attn_ab
 // Run:   ./attn_ab [rounds]
@@ -43,7 +45,9 @@ __global__ __launch_bounds__(512, 2) void attn_ab_kernel(
     constexpr int KVBLK = 64;
     constexpr int WAVES = 8;
     constexpr int SUPER = 1;
-    constexpr int DEPTH = (VAR == 1) ? 2 : 1;
+    constexpr int DEPTH = 1;
+    constexpr bool NT_IO = (VAR == 1);
+    constexpr bool STATIC_PRIO = (VAR == 1);
     constexpr int THREADS = WAVES * 64;
     constexpr int KPAD = D + 8;
     constexpr int VPAD = KVBLK + 8;
@@ -66,6 +70,9 @@ __global__ __launch_bounds__(512, 2) void attn_ab_kernel(
     const int wid = tid >> 6;
     const int l32 = lane & 31;
     const int hi = lane >> 5;
+    if (STATIC_PRIO &&
+        __builtin_amdgcn_readfirstlane(threadIdx.x) >= (WAVES / 2) * 64)
+        __builtin_amdgcn_s_setprio(1);
 
     // XCD-affine decode (1-D grid)
     const int nq = (S + WAVES * 32 - 1) / (WAVES * 32);
@@ -87,9 +94,11 @@ __global__ __launch_bounds__(512, 2) void attn_ab_kernel(
         const int row = q0 + l32;
         const int rr = row < S ? row : S - 1;
 #pragma unroll
-        for (int kk = 0; kk < KK; ++kk)
-            qfrag[kk] = *reinterpret_cast<const bf16x8*>(
+        for (int kk = 0; kk < KK; ++kk) {
+            const bf16x8* src = reinterpret_cast<const bf16x8*>(
                 qp + (long)rr * ss + kk * 16 + hi * 8);
+            qfrag[kk] = NT_IO ? __builtin_nontemporal_load(src) : *src;
+        }
     }
 
     f32x16 o_acc[NV];
@@ -302,9 +311,14 @@ __global__ __launch_bounds__(512, 2) void attn_ab_kernel(
                 for (int j = 0; j < 4; ++j)
                     pack[j] = __bfloat16_as_ushort(
                         f2bf(o_acc[n][r2 * 4 + j] * inv_l));
-                *reinterpret_cast<unsigned long long*>(
-                    op + (long)row * ss + dim0) =
+                unsigned long long* dst = reinterpret_cast<unsigned long long*>(
+                    op + (long)row * ss + dim0);
+                const unsigned long long pv =
                     *reinterpret_cast<unsigned long long*>(pack);
+                if (NT_IO)
+                    __builtin_nontemporal_store(pv, dst);
+                else
+                    *dst = pv;
             }
     }
 }
