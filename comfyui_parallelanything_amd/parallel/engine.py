@@ -309,6 +309,16 @@ class ParallelEngine:
             for ev in done_events:
                 if ev is not None:
                     lead_stream.wait_event(ev)
+            # order every worker stream after the lead-side copies so the
+            # caching allocator cannot reuse worker-side intermediates while
+            # a cross-device copy (which may have been enqueued on the lead
+            # stream) still reads them
+            ev_done = torch.cuda.Event()
+            ev_done.record(lead_stream)
+            for dev in devices:
+                s = self.streams.get(dev)
+                if s is not None:
+                    s.wait_event(ev_done)
         for i, r in enumerate(results):
             if r is None:
                 raise WorkerError(devices[i], RuntimeError("missing result"))
