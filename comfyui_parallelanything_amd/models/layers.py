@@ -210,8 +210,11 @@ class DoubleStreamBlock(nn.Module):
             self.img_attn_norm.key_norm.scale,
             pe,
         )
-        attn = ops.attention_bshd(q, k, v, self.attn.scale).flatten(2)
-        txt_attn, img_attn = attn[:, :T], attn[:, T:]
+        txt_attn, img_attn = ops.attention_bshd_split(
+            q, k, v, T, self.attn.scale
+        )
+        txt_attn = txt_attn.flatten(2)
+        img_attn = img_attn.flatten(2)
 
         img = ops.gate_residual(img, img_m1.gate, self.img_attn_proj(img_attn))
         img = ops.gate_residual(

@@ -209,6 +209,24 @@ def attention_bshd(q, k, v, scale: Optional[float] = None) -> torch.Tensor:
     return out.permute(0, 2, 1, 3)
 
 
+def attention_bshd_split(q, k, v, split: int, scale: Optional[float] = None):
+    """attention_bshd with per-stream outputs: rows [:split] and [split:]
+    land in two contiguous tensors (feeds dual-stream projections with no
+    reshape copies)."""
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    if q.is_cuda:
+        ext = _require_ext("attn_fwd_bshd_split")
+        if ext is not None and q.dtype == torch.bfloat16 \
+                and q.shape[-1] in (64, 128):
+            a, b = ext.attn_fwd_bshd_split(q, k, v, float(scale), int(split))
+            return a, b
+        _unsupported("attn_fwd_bshd_split",
+                     f"dtype={q.dtype}, D={q.shape[-1]}")
+    out = attention_bshd(q, k, v, scale)
+    return out[:, :split].contiguous(), out[:, split:].contiguous()
+
+
 def qk_norm_rope_(q, k, wq, wk, cs, eps: float = 1e-6):
     """In-place fused qk RMSNorm + RoPE on [B, S, H, D] views.
 

@@ -1094,7 +1094,9 @@ __global__ __launch_bounds__(512, 2) void attn_fwd_v4_kernel(
     long q_bs, long q_hs, int q_ss,
     long k_bs, long k_hs, int k_ss,
     long v_bs, long v_hs, int v_ss,
-    long o_bs, long o_hs, int o_ss, int xcd_grid) {
+    long o_bs, long o_hs, int o_ss, int xcd_grid,
+    bf16* __restrict__ out2, int s_split,
+    long o2_bs, long o2_hs, int o2_ss) {
     constexpr int KVBLK = 64;
     constexpr int WAVES = 8;
     constexpr int THREADS = WAVES * 64;
@@ -1137,6 +1139,7 @@ __global__ __launch_bounds__(512, 2) void attn_fwd_v4_kernel(
     const bf16* kp = k + b * k_bs + (long)h * k_hs;
     const bf16* vp = v + b * v_bs + (long)h * v_hs;
     bf16* op = out + b * o_bs + (long)h * o_hs;
+    bf16* op2 = out2 ? out2 + b * o2_bs + (long)h * o2_hs : nullptr;
 
     // Q fragments (B-operand of the swapped QK^T): lane holds
     // Q[q0 + l32][kk*16 + hi*8 + j], j = 0..7.
@@ -1293,6 +1296,21 @@ __global__ __launch_bounds__(512, 2) void attn_fwd_v4_kernel(
     (void)l_bcast;
     const int row = q0 + l32;
     if (row < S) {
+        // split-stream output: rows >= s_split land in out2 (per-stream
+        // contiguous buffers feed the txt/img projections with no reshape
+        // copies)
+        bf16* obase;
+        long orow;
+        int oss;
+        if (op2 != nullptr && row >= s_split) {
+            obase = op2;
+            orow = row - s_split;
+            oss = o2_ss;
+        } else {
+            obase = op;
+            orow = row;
+            oss = o_ss;
+        }
         const float inv_l = (l_run > 0.f) ? 1.f / l_run : 0.f;
 #pragma unroll
         for (int n = 0; n < NV; ++n) {
@@ -1306,7 +1324,7 @@ __global__ __launch_bounds__(512, 2) void attn_fwd_v4_kernel(
                     pack[j] = __bfloat16_as_ushort(
                         f2bf(o_acc[n][r2 * 4 + j] * inv_l));
                 *reinterpret_cast<unsigned long long*>(
-                    op + (long)row * o_ss + dim0) =
+                    obase + orow * oss + dim0) =
                     *reinterpret_cast<unsigned long long*>(pack);
             }
         }
@@ -1408,8 +1426,19 @@ static AttnStrides strides_of(const at::Tensor& t, int d_axis_check) {
     return AttnStrides{t.stride(0), t.stride(d_axis_check), (int)0};
 }
 
+static std::vector<at::Tensor> attn_fwd_launch_split(
+    at::Tensor q, at::Tensor k, at::Tensor v, double scale, bool bshd,
+    long split);
+
 static at::Tensor attn_fwd_launch(at::Tensor q, at::Tensor k, at::Tensor v,
                                   double scale, bool bshd) {
+    return attn_fwd_launch_split(q, k, v, scale, bshd, -1)[0];
+}
+
+
+static std::vector<at::Tensor> attn_fwd_launch_split(
+    at::Tensor q, at::Tensor k, at::Tensor v, double scale, bool bshd,
+    long split) {
     CHECK_GPU(q);
     TORCH_CHECK(q.scalar_type() == at::kBFloat16, "attn_fwd: bf16 only");
     TORCH_CHECK(q.dim() == 4, "attn_fwd expects 4-D q/k/v");
@@ -1431,9 +1460,22 @@ static at::Tensor attn_fwd_launch(at::Tensor q, at::Tensor k, at::Tensor v,
     const int Sk = (int)kc.size(s_ax);
     TORCH_CHECK((int)vc.size(s_ax) == Sk && (int)kc.size(h_ax) == H,
                 "attn: k/v shape mismatch");
+    const bool do_split = split > 0 && split < S && bshd;
     at::Tensor out = bshd
-        ? at::empty({B, S, H, D}, qc.options())
+        ? at::empty({B, do_split ? (int)split : S, H, D}, qc.options())
         : at::empty({B, H, S, D}, qc.options());
+    at::Tensor out2;
+    bf16* o2p = nullptr;
+    long o2_bs = 0, o2_hs = 0, ssplit = S;
+    int o2_ss = 0;
+    if (do_split) {
+        out2 = at::empty({B, S - (int)split, H, D}, qc.options());
+        o2p = (bf16*)out2.data_ptr();
+        o2_bs = out2.stride(0);
+        o2_hs = out2.stride(2);
+        o2_ss = (int)out2.stride(1);
+        ssplit = split;
+    }
     static const bool use_v3 = []() {
         const char* e = getenv("PA_ATTN_V3");
         return e && e[0] == '1';
@@ -1458,12 +1500,15 @@ static at::Tensor attn_fwd_launch(at::Tensor q, at::Tensor k, at::Tensor v,
         }
         if (D == 128) {
             hipLaunchKernelGGL(attn_fwd_v4_kernel<128>, grid_v4, dim3(512), 0,
-                               cur_stream(), PA_ATTN_ARGS, xcd_grid);
+                               cur_stream(), PA_ATTN_ARGS, xcd_grid,
+                               o2p, ssplit, o2_bs, o2_hs, o2_ss);
         } else {
             hipLaunchKernelGGL(attn_fwd_v4_kernel<64>, grid_v4, dim3(512), 0,
-                               cur_stream(), PA_ATTN_ARGS, xcd_grid);
+                               cur_stream(), PA_ATTN_ARGS, xcd_grid,
+                               o2p, ssplit, o2_bs, o2_hs, o2_ss);
         }
     } else if (D == 128) {
+        TORCH_CHECK(!do_split, "split output needs the v4 kernel");
         hipLaunchKernelGGL(attn_fwd_kernel<128>, grid, dim3(512), 0,
                            cur_stream(), PA_ATTN_ARGS);
     } else {
@@ -1471,7 +1516,8 @@ static at::Tensor attn_fwd_launch(at::Tensor q, at::Tensor k, at::Tensor v,
                            cur_stream(), PA_ATTN_ARGS);
     }
 #undef PA_ATTN_ARGS
-    return out;
+    if (do_split) return {out, out2};
+    return {out};
 }
 
 at::Tensor attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v, double scale) {
@@ -1480,6 +1526,12 @@ at::Tensor attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v, double scale) {
 
 at::Tensor attn_fwd_bshd(at::Tensor q, at::Tensor k, at::Tensor v, double scale) {
     return attn_fwd_launch(q, k, v, scale, /*bshd=*/true);
+}
+
+std::vector<at::Tensor> attn_fwd_bshd_split(at::Tensor q, at::Tensor k,
+                                            at::Tensor v, double scale,
+                                            long split) {
+    return attn_fwd_launch_split(q, k, v, scale, /*bshd=*/true, split);
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -1495,6 +1547,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("attn_fwd", &attn_fwd, "Fused flash attention fwd, bf16 MFMA (gfx950)");
     m.def("attn_fwd_bshd", &attn_fwd_bshd,
           "Fused flash attention fwd on [B,S,H,D] strided views (gfx950)");
+    m.def("attn_fwd_bshd_split", &attn_fwd_bshd_split,
+          "Attention with per-stream split outputs [B,:split]/[B,split:]");
     m.def("gelu_tanh", &gelu_tanh, "Vectorized tanh-GELU (gfx950)");
     m.def("pack_joint_qkv", &pack_joint_qkv,
           "Fused dual-stream qkv pack + qk-norm + RoPE (gfx950)");

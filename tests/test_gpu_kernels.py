@@ -298,3 +298,20 @@ def test_attn_v3_fallback_env():
     r = subprocess.run([sys.executable, "-c", code], env=env,
                        cwd=os.path.dirname(os.path.dirname(__file__)))
     assert r.returncode == 0
+
+
+def test_attn_split_outputs():
+    torch.manual_seed(11)
+    B, S, H, D = 2, 192, 8, 128  # BH=16 (XCD grid) with split mid-sequence
+    q = torch.randn(B, S, H, D, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn_like(q)
+    v = torch.randn_like(q)
+    a, b = ops.attention_bshd_split(q, k, v, 64)
+    full = ops.attention_bshd(q, k, v)
+    _cmp(a, full[:, :64], 0, 0, "split txt part")  # same kernel math: bitwise
+    _cmp(b, full[:, 64:], 0, 0, "split img part")
+    ref = R.attention(
+        q.permute(0, 2, 1, 3).float(), k.permute(0, 2, 1, 3).float(),
+        v.permute(0, 2, 1, 3).float(),
+    ).permute(0, 2, 1, 3)
+    _cmp(torch.cat([a, b], dim=1), ref, 2e-2, 2e-2, "split vs reference")
