@@ -81,3 +81,29 @@ def test_samplers_with_tiny_model():
     for name in ("euler", "heun"):
         out = SAMPLERS[name](m, x.clone(), flow_sigmas(3), context=c, **kw)
         assert out.shape == x.shape and torch.isfinite(out).all()
+
+
+def test_end_to_end_denoise_with_engine():
+    """Full sampler loop through the parallel-installed model on [cpu,cpu]
+    matches the single-device loop exactly (BASELINE config 1, end to end)."""
+    from comfyui_parallelanything_amd.models.registry import make_sd15, sd15_inputs
+    from comfyui_parallelanything_amd.parallel.chain import DeviceChain, make_entry
+    from comfyui_parallelanything_amd.parallel.engine import (
+        ParallelEngine,
+        install_parallel_forward,
+    )
+    from comfyui_parallelanything_amd.sampling import karras_sigmas, sample_dpmpp_2m
+
+    m = make_sd15(tiny=True)
+    x, t, c, kw = sd15_inputs(2, tiny=True)
+    sig = karras_sigmas(6)
+    ref = sample_dpmpp_2m(m, x.clone(), sig, context=c, **kw)
+
+    eng = ParallelEngine(
+        DeviceChain.from_list([make_entry("cpu", 50), make_entry("cpu", 50)]),
+        auto_vram_balance=False,
+    )
+    eng.setup(m)
+    install_parallel_forward(m, eng)
+    out = sample_dpmpp_2m(m, x.clone(), sig, context=c, **kw)
+    torch.testing.assert_close(out, ref, rtol=1e-4, atol=1e-5)
