@@ -82,6 +82,9 @@ def main():
         )
     dev = info.device
 
+    if args.model in ("sd15", "sdxl") and on_gpu:
+        # MIOpen conv algo search (cached in-process) for the UNet families
+        torch.backends.cudnn.benchmark = True
     make, make_inputs = MODELS[args.model]
     model = make(dev=dev, dtype=dtype, tiny=tiny)
     # replicate(): rank0's weights to every replica, flat bucketed RCCL bcast
