@@ -61,3 +61,24 @@ def _zimage_worker(rank, world, tmpdir):
 @pytest.mark.parametrize("worker", [_flux_worker, _zimage_worker])
 def test_dist_pipeline_world2(worker, tmp_path):
     mp.spawn(worker, args=(2, str(tmp_path)), nprocs=2, join=True)
+
+
+def _flux_worker3(rank, world, tmpdir):
+    from comfyui_parallelanything_amd.models.registry import flux_inputs, make_flux
+    from comfyui_parallelanything_amd.parallel.dist_pipeline import (
+        install_dist_pipeline,
+    )
+
+    info = _init(rank, world, tmpdir)
+    model = make_flux(tiny=True, dtype=torch.float32)
+    x, t, c, kw = flux_inputs(1, tiny=True, dtype=torch.float32)
+    ref = model(x, t, context=c, **kw)
+    install_dist_pipeline(model, info, weights=[0.5, 0.3, 0.2])
+    out = model(x, t, context=c, **kw)
+    if rank == 0:
+        torch.testing.assert_close(out, ref, rtol=1e-5, atol=1e-6)
+    dist.destroy_process_group()
+
+
+def test_dist_pipeline_world3(tmp_path):
+    mp.spawn(_flux_worker3, args=(3, str(tmp_path)), nprocs=3, join=True)
