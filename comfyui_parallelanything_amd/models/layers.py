@@ -201,18 +201,18 @@ class DoubleStreamBlock(nn.Module):
         txt_qkv = self.txt_attn_qkv(txt_in).unflatten(-1, (3, H, -1))
         img_qkv = self.img_attn_qkv(img_in).unflatten(-1, (3, H, -1))
         # fused: per-stream qk-norm + RoPE + contiguous joint q/k/v
-        # (txt first, then img — FLUX convention). The 1/sqrt(D) softmax
-        # scale is folded into the q-norm weight (256 B op) so the
-        # attention kernel skips 32 VALU multiplies per KV tile.
+        # (txt first, then img — FLUX convention)
         q, k, v = ops.pack_joint_qkv(
             txt_qkv, img_qkv,
-            self.txt_attn_norm.query_norm.scale * self.attn.scale,
+            self.txt_attn_norm.query_norm.scale,
             self.txt_attn_norm.key_norm.scale,
-            self.img_attn_norm.query_norm.scale * self.attn.scale,
+            self.img_attn_norm.query_norm.scale,
             self.img_attn_norm.key_norm.scale,
             pe,
         )
-        txt_attn, img_attn = ops.attention_bshd_split(q, k, v, T, 1.0)
+        txt_attn, img_attn = ops.attention_bshd_split(
+            q, k, v, T, self.attn.scale
+        )
         txt_attn = txt_attn.flatten(2)
         img_attn = img_attn.flatten(2)
 
@@ -258,12 +258,10 @@ class SingleStreamBlock(nn.Module):
         qkv = self.linear1_qkv(x_in).unflatten(-1, (3, self.num_heads, -1))
         mlp_act = self.linear1_mlp(x_in)  # GELU in the GEMM epilogue
         q, k, v = qkv.unbind(2)  # [B,S,H,D] views
-        # softmax scale folded into the q-norm weight (see DoubleStreamBlock)
         ops.qk_norm_rope_(
-            q, k, self.norm.query_norm.scale * self.attn.scale,
-            self.norm.key_norm.scale, pe,
+            q, k, self.norm.query_norm.scale, self.norm.key_norm.scale, pe
         )
-        attn = ops.attention_bshd(q, k, v, 1.0).flatten(2)
+        attn = ops.attention_bshd(q, k, v, self.attn.scale).flatten(2)
         if x.is_cuda and isinstance(self.linear2_mlp, nn.Linear):
             # second GEMM accumulates into the first's output (beta=1
             # epilogue) — no separate elementwise add

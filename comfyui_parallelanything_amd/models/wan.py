@@ -58,10 +58,10 @@ class WanBlock(nn.Module):
         qkv = self.self_qkv(h).unflatten(-1, (3, self.num_heads, -1))
         q, k, v = qkv.unbind(2)  # [B,S,H,D] views
         ops.qk_norm_rope_(
-            q, k, self.self_norm.query_norm.scale * self.scale,
+            q, k, self.self_norm.query_norm.scale,
             self.self_norm.key_norm.scale, pe,
         )
-        attn = ops.attention_bshd(q, k, v, 1.0).flatten(2)
+        attn = ops.attention_bshd(q, k, v, self.scale).flatten(2)
         x = ops.gate_residual(x, gate1, self.self_proj(attn))
 
         h = self.norm_cross(x)
