@@ -1224,18 +1224,21 @@ __global__ __launch_bounds__(512, 2) void attn_fwd_v4_kernel(
 
         // ---- lane-local online softmax (this lane's row = q0 + l32) ------
         // value (kt, reg) = S[row][key = kt*32 + (reg&3) + 8*(reg>>2) + 4*hi]
-        float mx = -1e30f;
+        // track the RAW tile max (scale2 > 0 commutes with max) and fuse
+        // the softmax scale into the exp2 argument as one FMA per element —
+        // saves a 32-wide VALU multiply pass per tile.
+        float mx = -3e30f;
 #pragma unroll
         for (int kt = 0; kt < 2; ++kt)
 #pragma unroll
             for (int r = 0; r < 16; ++r) {
                 const int key = kv0 + kt * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
-                const float sv = (key < Sk) ? st[kt][r] * scale2 : -1e30f;
+                const float sv = (key < Sk) ? st[kt][r] : -3e30f;
                 st[kt][r] = sv;
                 mx = fmaxf(mx, sv);
             }
         mx = fmaxf(mx, __shfl_xor(mx, 32, 64));  // partner holds the row's other 32 keys
-        const float mnew = fmaxf(m_run, mx);
+        const float mnew = fmaxf(m_run, mx * scale2);
         const float alpha = __builtin_amdgcn_exp2f(m_run - mnew);
         m_run = mnew;
         float ps = 0.f;
@@ -1243,7 +1246,8 @@ __global__ __launch_bounds__(512, 2) void attn_fwd_v4_kernel(
         for (int kt = 0; kt < 2; ++kt)
 #pragma unroll
             for (int r = 0; r < 16; ++r) {
-                const float pv_ = __builtin_amdgcn_exp2f(st[kt][r] - mnew);
+                const float pv_ =
+                    __builtin_amdgcn_exp2f(fmaf(st[kt][r], scale2, -mnew));
                 st[kt][r] = pv_;
                 ps += pv_;
             }
