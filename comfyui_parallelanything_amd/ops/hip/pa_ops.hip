@@ -1108,7 +1108,6 @@ __global__ __launch_bounds__(512, 2) void attn_fwd_v4_kernel(
 
     __shared__ bf16 k_lds[KVBLK * KPAD];
     __shared__ bf16 v_lds[D * VPAD];
-    __shared__ float l_bcast[WAVES];  // unused slot keeps LDS struct simple
 
     const int tid = threadIdx.x;
     const int lane = tid & 63;
@@ -1297,7 +1296,6 @@ __global__ __launch_bounds__(512, 2) void attn_fwd_v4_kernel(
     }
 
     // ---- epilogue: O = O^T / l, row is lane-local -------------------------
-    (void)l_bcast;
     const int row = q0 + l32;
     if (row < S) {
         // split-stream output: rows >= s_split land in out2 (per-stream
