@@ -52,6 +52,11 @@ class FP8Linear(nn.Module):
         self.out_dtype = out_dtype
         self.in_features = weight_fp8.shape[1]
         self.out_features = weight_fp8.shape[0]
+        # delayed activation scaling: running amax updated by the fused
+        # quant kernel each call; the scale used is the PREVIOUS call's
+        self.register_buffer("a_amax", torch.zeros(1, dtype=torch.float32),
+                             persistent=False)
+        self._warm = False
 
     @classmethod
     def from_linear(cls, lin: nn.Linear) -> "FP8Linear":
@@ -64,13 +69,25 @@ class FP8Linear(nn.Module):
         return cls(w8, w_scale.to(w.device), bias, lin.weight.dtype)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        from .. import ops
+
         shape = x.shape
         x2 = x.reshape(-1, shape[-1])
-        x_scale = (x2.abs().amax().float() / FP8_MAX).clamp(min=1e-12)
-        x8 = (x2.float() / x_scale).clamp(-FP8_MAX, FP8_MAX).to(FP8)
+        if (x.is_cuda and x.dtype == torch.bfloat16
+                and ops.hip_available("quant_fp8")):
+            if not self._warm:
+                # first call: measure directly (still async, device-side)
+                self.a_amax.copy_(x2.abs().amax().reshape(1).float())
+                self._warm = True
+            x_scale = (self.a_amax / FP8_MAX).clamp(min=1e-12)
+            self.a_amax.mul_(0.999)  # slow decay lets the scale shrink
+            x8 = ops.quant_fp8(x2, x_scale, self.a_amax)
+        else:
+            x_scale = (x2.abs().amax().float() / FP8_MAX).clamp(min=1e-12)
+            x8 = (x2.float() / x_scale).clamp(-FP8_MAX, FP8_MAX).to(FP8)
         y = torch._scaled_mm(
             x8, self.weight_fp8.t(),
-            scale_a=x_scale, scale_b=self.w_scale,
+            scale_a=x_scale.reshape(()), scale_b=self.w_scale,
             bias=self.bias.to(self.out_dtype) if self.bias is not None else None,
             out_dtype=self.out_dtype,
         )

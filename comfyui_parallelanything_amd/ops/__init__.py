@@ -268,6 +268,14 @@ def pack_joint_qkv(txt_qkv, img_qkv, wq_t, wk_t, wq_i, wk_i, cs,
     return q, k, v
 
 
+def quant_fp8(x: torch.Tensor, scale: torch.Tensor,
+              amax_buf: torch.Tensor) -> torch.Tensor:
+    """Fused bf16 -> e4m3fn cast with running-amax update (delayed scaling).
+    GPU-only; used by the fp8 serving mode."""
+    ext = _require_ext("quant_fp8")
+    return ext.quant_fp8(x, scale, amax_buf)
+
+
 def gelu_tanh(x: torch.Tensor) -> torch.Tensor:
     if x.is_cuda:
         ext = _require_ext("gelu_tanh")

@@ -35,3 +35,37 @@ def test_quantize_fp8_model_runs():
     with torch.no_grad():
         out = m(x, t, context=c, **kw)
     assert torch.isfinite(out.float()).all()
+
+
+def test_quant_fp8_kernel():
+    from comfyui_parallelanything_amd import ops
+
+    if not ops.hip_available("quant_fp8"):
+        pytest.skip("no quant_fp8 in extension")
+    torch.manual_seed(0)
+    x = torch.randn(256, 1024, device="cuda", dtype=torch.bfloat16) * 3
+    scale = torch.tensor([x.abs().amax().item() / 448.0], device="cuda")
+    amax = torch.zeros(1, device="cuda")
+    x8 = ops.quant_fp8(x, scale, amax)
+    assert x8.dtype == torch.float8_e4m3fn
+    ref = (x.float() / scale).clamp(-448, 448).to(torch.float8_e4m3fn)
+    # compare dequantized values
+    torch.testing.assert_close(x8.float(), ref.float(), rtol=0, atol=0)
+    # fused amax matches the true amax
+    assert abs(amax.item() - x.abs().amax().item()) < 1e-3
+
+
+def test_fp8_linear_delayed_scaling_stable():
+    from comfyui_parallelanything_amd.models.quant import FP8Linear, _supports_scaled_mm
+
+    if not _supports_scaled_mm():
+        pytest.skip("no fp8 _scaled_mm")
+    torch.manual_seed(1)
+    lin = torch.nn.Linear(1024, 1024).cuda().to(torch.bfloat16)
+    q = FP8Linear.from_linear(lin)
+    x = torch.randn(64, 1024, device="cuda", dtype=torch.bfloat16)
+    ref = lin(x).float()
+    outs = [q(x).float() for _ in range(4)]  # delayed scale warm + steady
+    for out in outs[1:]:
+        rel = (out - ref).norm() / ref.norm()
+        assert rel < 0.06, f"fp8 delayed-scale error {rel:.4f}"
