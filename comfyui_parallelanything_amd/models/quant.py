@@ -53,9 +53,14 @@ class FP8Linear(nn.Module):
         self.in_features = weight_fp8.shape[1]
         self.out_features = weight_fp8.shape[0]
         # delayed activation scaling: running amax updated by the fused
-        # quant kernel each call; the scale used is the PREVIOUS call's
-        self.register_buffer("a_amax", torch.zeros(1, dtype=torch.float32),
-                             persistent=False)
+        # quant kernel each call; the scale used is the PREVIOUS call's.
+        # MUST live on the weight's device — the module is swapped in place
+        # and never .to()-ed afterwards.
+        self.register_buffer(
+            "a_amax",
+            torch.zeros(1, dtype=torch.float32, device=weight_fp8.device),
+            persistent=False,
+        )
         self._warm = False
 
     @classmethod
