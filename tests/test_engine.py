@@ -126,3 +126,27 @@ def test_install_uninstall_roundtrip(sd15):
     uninstall_parallel_forward(sd15)
     assert not hasattr(sd15, "_true_parallel_active")
     assert sd15.forward == orig_forward
+
+
+def test_forward_without_context(sd15):
+    x, t, _, _ = sd15_inputs(2, tiny=True)
+    eng = ParallelEngine(cpu_chain(50, 50), auto_vram_balance=False)
+    eng.setup(sd15)
+    out = eng.forward(x, t)  # context omitted entirely
+    assert torch.equal(out, sd15(x, t))
+
+
+def test_release_clears_state(sd15):
+    eng = ParallelEngine(cpu_chain(50, 50), auto_vram_balance=False)
+    eng.setup(sd15)
+    assert eng.replicas
+    eng.release()
+    assert not eng.replicas and not eng.streams and eng.pipeline is None
+
+
+def test_single_device_chain_routes_lead_only(sd15):
+    x, t, c, kw = sd15_inputs(4, tiny=True)
+    eng = ParallelEngine(cpu_chain(100), auto_vram_balance=False)
+    eng.setup(sd15)
+    out = eng.forward(x, t, context=c, **kw)
+    assert torch.equal(out, sd15(x, t, context=c, **kw))
