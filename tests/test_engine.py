@@ -133,7 +133,7 @@ def test_forward_without_context(sd15):
     eng = ParallelEngine(cpu_chain(50, 50), auto_vram_balance=False)
     eng.setup(sd15)
     out = eng.forward(x, t)  # context omitted entirely
-    assert torch.equal(out, sd15(x, t))
+    torch.testing.assert_close(out, sd15(x, t), rtol=1e-4, atol=1e-5)
 
 
 def test_release_clears_state(sd15):
