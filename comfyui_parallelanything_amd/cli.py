@@ -39,6 +39,8 @@ def main(argv=None):
     ap.add_argument("--tiny", action="store_true")
     ap.add_argument("--no-split", dest="split", action="store_false")
     ap.add_argument("--no-balance", dest="balance", action="store_false")
+    ap.add_argument("--microbatches", type=int, default=1,
+                    help="pipeline-mode micro-batches (>1 overlaps stages)")
     args = ap.parse_args(argv)
 
     devices = args.devices.split(",")
@@ -64,7 +66,7 @@ def main(argv=None):
     engine = ParallelEngine(chain, workload_split=args.split,
                             auto_vram_balance=args.balance)
     engine.setup(model)
-    configure_pipeline(engine)
+    configure_pipeline(engine, microbatches=args.microbatches)
     install_parallel_forward(model, engine)
 
     if args.model == "wan":

@@ -122,6 +122,11 @@ class ParallelAnything:
                 "purge_models": ("BOOLEAN", {
                     "default": False,
                     "tooltip": "Unload all models when cleaning up (aggressive)"}),
+                "pipeline_microbatches": ("INT", {
+                    "default": 1, "min": 1, "max": 16,
+                    "tooltip": "Micro-batches for block-sharded pipeline mode "
+                               "(>1 overlaps stages for small batches; "
+                               "extension over the reference)"}),
             },
         }
 
@@ -131,7 +136,8 @@ class ParallelAnything:
     CATEGORY = "utils/hardware"
 
     def setup_parallel(self, model, device_chain, workload_split=True,
-                       auto_vram_balance=True, purge_cache=True, purge_models=False):
+                       auto_vram_balance=True, purge_cache=True,
+                       purge_models=False, pipeline_microbatches=1):
         if model is None or not device_chain:
             return (model,)
 
@@ -172,7 +178,7 @@ class ParallelAnything:
             log.exception("replication failed; returning model unchanged")
             engine.release()
             return (model,)
-        configure_pipeline(engine)
+        configure_pipeline(engine, microbatches=pipeline_microbatches)
 
         install_parallel_forward(target_model, engine)
         target_model._parallel_purge_cache = purge_cache

@@ -152,8 +152,17 @@ class ParallelEngine:
         batch = get_batch_size(x)
         devices, weights = list(self.chain.devices), list(self.chain.weights)
 
-        # Routing ladder (reference :1295-1346).
-        if batch == 1 and self.workload_split and self.pipeline is not None:
+        # Routing ladder (reference :1295-1346). Batches too small for DP
+        # also take the pipeline when micro-batching is enabled (extension;
+        # the reference sends 1 < batch < n_devices to the lead only).
+        if (
+            self.workload_split
+            and self.pipeline is not None
+            and (
+                batch == 1
+                or (batch < len(devices) and self.pipeline.microbatches > 1)
+            )
+        ):
             return self.pipeline.forward(x, timesteps, context, **kwargs)
         if not self.workload_split or batch < len(devices) or len(devices) == 1:
             return self._lead_only(x, timesteps, context, **kwargs)

@@ -117,10 +117,21 @@ def assign_block_ranges(num_blocks: int, weights) -> List[int]:
 
 class PipelineConfig:
     """Wires ParallelBlock wrappers into the lead replica and exposes the
-    batch==1 forward (engine routes here; reference :1295-1305)."""
+    batch==1 forward (engine routes here; reference :1295-1305).
 
-    def __init__(self, engine):
+    ``microbatches`` > 1 enables GPipe-style micro-batching (NOT in the
+    reference — SURVEY.md §6 flags it as an optional extension): a batch
+    B > 1 that is too small for DP is split into ``min(microbatches, B)``
+    micro-batches, each driven through the block-sharded pipeline by its
+    own host thread. Per device, micro-batch work serializes on that
+    device's current stream (correct stage ordering for free); ACROSS
+    devices the stages overlap, so stage 1 processes micro-batch m while
+    stage 0 runs m+1 — classic fill/drain. Inference-only (no_grad),
+    stateless block bodies, so concurrent threads are safe."""
+
+    def __init__(self, engine, microbatches: int = 1):
         self.engine = engine
+        self.microbatches = max(1, int(microbatches))
         self.configured = False
         self._wire()
 
@@ -153,19 +164,63 @@ class PipelineConfig:
 
     @torch.no_grad()
     def forward(self, x, timesteps, context=None, **kwargs):
+        from .split import (
+            concatenate_results,
+            get_batch_size,
+            split_batch,
+            split_kwargs,
+        )
+
         eng = self.engine
-        set_pipeline_mode(True)
-        try:
-            return eng._lead_only(x, timesteps, context, **kwargs)
-        finally:
-            set_pipeline_mode(False)
+        batch = get_batch_size(x)
+        n_mb = min(self.microbatches, batch)
+        if n_mb <= 1:
+            set_pipeline_mode(True)
+            try:
+                return eng._lead_only(x, timesteps, context, **kwargs)
+            finally:
+                set_pipeline_mode(False)
+
+        sizes = [batch // n_mb + (1 if i < batch % n_mb else 0)
+                 for i in range(n_mb)]
+        x_mb = split_batch(x, sizes)
+        t_mb = split_batch(timesteps, sizes)
+        c_mb = split_batch(context, sizes) if context is not None else None
+        kw_mb = split_kwargs(kwargs, sizes, batch)
+
+        results: List = [None] * n_mb
+        errors: List = []
+
+        def run(i: int) -> None:
+            set_pipeline_mode(True)  # flag is thread-local: set per thread
+            try:
+                with torch.no_grad():
+                    results[i] = eng._lead_only(
+                        x_mb[i], t_mb[i],
+                        c_mb[i] if c_mb is not None else None,
+                        **kw_mb[i],
+                    )
+            except BaseException as err:  # noqa: BLE001
+                errors.append(err)
+            finally:
+                set_pipeline_mode(False)
+
+        threads = [threading.Thread(target=run, args=(i,), daemon=True)
+                   for i in range(n_mb)]
+        for t in threads:  # launch in order: micro-batch m fills stage 0 first
+            t.start()
+        for t in threads:
+            t.join()
+        if errors:
+            raise errors[0]
+        return concatenate_results(results, dim=0)
 
 
-def configure_pipeline(engine) -> None:
+def configure_pipeline(engine, microbatches: int = 1) -> None:
     """Attach pipeline mode to an engine when >1 device and the model
     exposes a recognized block list; silently a no-op otherwise."""
     if len(engine.chain.devices) < 2:
         return
-    cfg = PipelineConfig(engine)
+    cfg = PipelineConfig(engine, microbatches=microbatches)
     if cfg.configured:
         engine.pipeline = cfg

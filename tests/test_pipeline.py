@@ -79,6 +79,71 @@ def test_unwrap_restores_blocks():
     assert not any(isinstance(b, ParallelBlock) for b in lead.double_blocks)
 
 
+def test_microbatched_pipeline_matches_reference():
+    """GPipe-style micro-batching: batch 3 through a 3-device pipeline
+    in 3 concurrent micro-batches equals the single-device forward."""
+    m = make_flux(tiny=True, dtype=torch.float32)
+    eng = ParallelEngine(cpu_chain(34, 33, 33), auto_vram_balance=False)
+    eng.setup(m, force_copy_lead=True)
+    configure_pipeline(eng, microbatches=4)
+    x, t, c, kw = flux_inputs(3, tiny=True, dtype=torch.float32)
+    ref = m(x, t, context=c, **kw)
+    out = eng.forward(x, t, context=c, **kw)  # 3 < 3 devices? no: 3 == 3 → DP
+    torch.testing.assert_close(out, ref, rtol=1e-4, atol=1e-5)
+    # batch 2 < 3 devices → pipeline with 2 micro-batches (DP impossible)
+    x, t, c, kw = flux_inputs(2, tiny=True, dtype=torch.float32)
+    ref = m(x, t, context=c, **kw)
+    out = eng.pipeline.forward(x, t, context=c, **kw)
+    torch.testing.assert_close(out, ref, rtol=1e-4, atol=1e-5)
+    assert not pipeline_mode_active()
+
+
+def test_microbatch_routing_small_batch():
+    """1 < batch < n_devices routes to the pipeline when micro-batching is
+    on, and to the lead when it is off (reference behavior)."""
+    m = make_flux(tiny=True, dtype=torch.float32)
+    eng = ParallelEngine(cpu_chain(34, 33, 33), auto_vram_balance=False)
+    eng.setup(m, force_copy_lead=True)
+    configure_pipeline(eng, microbatches=2)
+
+    calls = {"pipe": 0}
+    orig = eng.pipeline.forward
+
+    def spy(*a, **kw):
+        calls["pipe"] += 1
+        return orig(*a, **kw)
+
+    eng.pipeline.forward = spy
+    x, t, c, kw = flux_inputs(2, tiny=True, dtype=torch.float32)
+    ref = m(x, t, context=c, **kw)
+    out = eng.forward(x, t, context=c, **kw)
+    assert calls["pipe"] == 1
+    torch.testing.assert_close(out, ref, rtol=1e-4, atol=1e-5)
+
+    eng.pipeline.microbatches = 1  # reference routing: lead-only
+    eng.forward(x, t, context=c, **kw)
+    assert calls["pipe"] == 1
+
+
+def test_microbatch_error_propagates_and_clears_flag():
+    class Bad(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.layers = torch.nn.ModuleList([torch.nn.Identity()])
+
+        def forward(self, x, t, context=None):
+            raise ValueError("mb failure")
+
+    eng = ParallelEngine(cpu_chain(50, 50), auto_vram_balance=False)
+    eng.setup(Bad(), force_copy_lead=True)
+    configure_pipeline(eng, microbatches=2)
+    if eng.pipeline is None:
+        pytest.skip("no block list wired")
+    with pytest.raises(ValueError):
+        eng.pipeline.forward(torch.zeros(4, 3), torch.zeros(4))
+    assert not pipeline_mode_active()
+
+
 def test_flag_cleared_on_error():
     class Bad(torch.nn.Module):
         def __init__(self):
