@@ -127,6 +127,11 @@ class ParallelAnything:
                     "tooltip": "Micro-batches for block-sharded pipeline mode "
                                "(>1 overlaps stages for small batches; "
                                "extension over the reference)"}),
+                "use_hip_graphs": ("BOOLEAN", {
+                    "default": False,
+                    "tooltip": "Capture repeated same-shape forwards into "
+                               "hipGraphs (one launch per denoise step; "
+                               "extension over the reference)"}),
             },
         }
 
@@ -137,7 +142,8 @@ class ParallelAnything:
 
     def setup_parallel(self, model, device_chain, workload_split=True,
                        auto_vram_balance=True, purge_cache=True,
-                       purge_models=False, pipeline_microbatches=1):
+                       purge_models=False, pipeline_microbatches=1,
+                       use_hip_graphs=False):
         if model is None or not device_chain:
             return (model,)
 
@@ -171,6 +177,7 @@ class ParallelAnything:
             chain,
             workload_split=workload_split,
             auto_vram_balance=auto_vram_balance,
+            use_hip_graphs=use_hip_graphs,
         )
         try:
             engine.setup(target_model, force_copy_lead=has_lora)

@@ -75,6 +75,7 @@ class ParallelEngine:
         chain: DeviceChain,
         workload_split: bool = True,
         auto_vram_balance: bool = True,
+        use_hip_graphs: bool = False,
     ):
         self.chain = chain
         self.workload_split = workload_split
@@ -82,6 +83,13 @@ class ParallelEngine:
         self.replicas: Dict[str, nn.Module] = {}
         self.streams: Dict[str, Optional[torch.cuda.Stream]] = {}
         self.pipeline = None  # set by pipeline.configure_pipeline
+        # shape-keyed hipGraph capture of per-replica forwards (extension;
+        # see hipgraphs.py). Off by default: eager == reference behavior.
+        self.graphs = None
+        if use_hip_graphs:
+            from .hipgraphs import GraphRunner
+
+            self.graphs = GraphRunner()
         self._lead_t = torch.device(chain.lead)
         # per-step timing feedback (closes the reference's static-balancing
         # limitation); engaged only when auto_vram_balance is on
@@ -367,6 +375,13 @@ class ParallelEngine:
         # aliasing, reference :594-597): call through _original_forward to
         # avoid re-entering the scheduler (reference :1390).
         fwd = getattr(replica, "_original_forward", None) or replica
+        if self.graphs is not None:
+            from .pipeline import pipeline_mode_active
+
+            # pipeline mode crosses devices inside the forward: a single
+            # per-device graph cannot capture that — stay eager there.
+            if not pipeline_mode_active():
+                return self.graphs.run(fwd, dev, xi, ti, ci, kwi)
         if ci is not None:
             return fwd(xi, ti, context=ci, **kwi)
         return fwd(xi, ti, **kwi)
@@ -378,6 +393,8 @@ class ParallelEngine:
         self.replicas.clear()
         self.streams.clear()
         self.pipeline = None
+        if self.graphs is not None:
+            self.graphs.clear()
         if torch.cuda.is_available():
             torch.cuda.empty_cache()
 

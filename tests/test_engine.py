@@ -144,6 +144,39 @@ def test_release_clears_state(sd15):
     assert not eng.replicas and not eng.streams and eng.pipeline is None
 
 
+def test_hip_graphs_cpu_falls_back_to_eager(sd15):
+    """use_hip_graphs on a CPU chain must be a transparent no-op."""
+    x, t, c, kw = sd15_inputs(2, tiny=True)
+    ref = sd15(x, t, context=c, **kw)
+    eng = ParallelEngine(
+        cpu_chain(100), auto_vram_balance=False, use_hip_graphs=True
+    )
+    eng.setup(sd15)
+    for _ in range(4):  # past WARMUP_CALLS: still must not try to capture
+        out = eng.forward(x, t, context=c, **kw)
+    assert torch.equal(out, ref)
+    assert not eng.graphs._graphs
+    eng.release()
+
+
+def test_graph_runner_key_rules():
+    from comfyui_parallelanything_amd.parallel.hipgraphs import GraphRunner
+
+    r = GraphRunner()
+    x, t = torch.zeros(2, 3), torch.zeros(2)
+    k1 = r.key_for("cuda:0", x, t, None, {"y": torch.ones(2, 5)})
+    k2 = r.key_for("cuda:0", x, t, None, {"y": torch.ones(2, 5)})
+    assert k1 == k2 and k1 is not None
+    # different shape, device, or dtype -> different signature
+    assert r.key_for("cuda:0", torch.zeros(3, 3), t, None, {}) != k1
+    assert r.key_for("cuda:1", x, t, None, {"y": torch.ones(2, 5)}) != k1
+    assert (
+        r.key_for("cuda:0", x.double(), t, None, {"y": torch.ones(2, 5)}) != k1
+    )
+    # non-tensor kwarg -> not graphable
+    assert r.key_for("cuda:0", x, t, None, {"flag": True}) is None
+
+
 def test_single_device_chain_routes_lead_only(sd15):
     x, t, c, kw = sd15_inputs(4, tiny=True)
     eng = ParallelEngine(cpu_chain(100), auto_vram_balance=False)

@@ -81,6 +81,36 @@ def test_install_forward_cleanup_gpu():
     assert not getattr(m, "_true_parallel_active", False)
 
 
+def test_hip_graph_engine_matches_eager():
+    """use_hip_graphs: first two calls are eager warmup, third captures,
+    later calls replay — all must match the eager engine bit-for-bit-ish
+    (same kernels, same order -> tight tolerance)."""
+    m = make_flux(dev="cuda:0", dtype=torch.bfloat16, tiny=True)
+    x, t, c, kw = flux_inputs(2, dev="cuda:0", dtype=torch.bfloat16, tiny=True)
+    eager = ParallelEngine(chain(("cuda:0", 100)), auto_vram_balance=False)
+    eager.setup(m)
+    ref = eager.forward(x, t, context=c, **kw).clone()
+
+    eng = ParallelEngine(
+        chain(("cuda:0", 100)), auto_vram_balance=False, use_hip_graphs=True
+    )
+    eng.setup(m)
+    for i in range(5):
+        out = eng.forward(x, t, context=c, **kw)
+        torch.cuda.synchronize()
+        torch.testing.assert_close(out, ref, rtol=2e-2, atol=2e-2), i
+    assert eng.graphs._graphs, "no hipGraph was captured after warmup"
+    # changed shape -> new signature, falls back to eager warmup, still right
+    x3, t3, c3, kw3 = flux_inputs(
+        3, dev="cuda:0", dtype=torch.bfloat16, tiny=True
+    )
+    out3 = eng.forward(x3, t3, context=c3, **kw3)
+    assert out3.shape == x3.shape
+    assert torch.isfinite(out3.float()).all()
+    eng.release()
+    assert not eng.graphs._graphs
+
+
 def test_vram_balancer_reads_hbm():
     from comfyui_parallelanything_amd.parallel.balance import get_free_vram_mb
 
