@@ -2,6 +2,10 @@
 
 Evidence run for engine.use_hip_graphs (see parallel/hipgraphs.py).
 """
+import sys
+
+sys.path.insert(0, __import__("os").path.dirname(__import__("os").path.dirname(__import__("os").path.abspath(__file__))))
+
 import time
 
 import torch
