@@ -41,6 +41,8 @@ def main(argv=None):
     ap.add_argument("--no-balance", dest="balance", action="store_false")
     ap.add_argument("--microbatches", type=int, default=1,
                     help="pipeline-mode micro-batches (>1 overlaps stages)")
+    ap.add_argument("--hip-graphs", action="store_true",
+                    help="capture repeated same-shape forwards into hipGraphs")
     args = ap.parse_args(argv)
 
     devices = args.devices.split(",")
@@ -64,7 +66,8 @@ def main(argv=None):
     model = make(dev=chain.lead, dtype=dtype, tiny=tiny)
 
     engine = ParallelEngine(chain, workload_split=args.split,
-                            auto_vram_balance=args.balance)
+                            auto_vram_balance=args.balance,
+                            use_hip_graphs=args.hip_graphs)
     engine.setup(model)
     configure_pipeline(engine, microbatches=args.microbatches)
     install_parallel_forward(model, engine)

@@ -69,6 +69,24 @@ def test_pipeline_mode_gpu_lead():
     eng.release()
 
 
+def test_microbatched_pipeline_gpu_lead():
+    """batch 2 < 3 devices -> pipeline with 2 concurrent micro-batches on a
+    [cuda, cuda, cpu] hybrid chain; matches the single-device forward."""
+    m = make_flux(dev="cuda:0", dtype=torch.float32, tiny=True)
+    x, t, c, kw = flux_inputs(2, dev="cuda:0", dtype=torch.float32, tiny=True)
+    ref = m(x, t, context=c, **kw)
+    eng = ParallelEngine(
+        chain(("cuda:0", 40), ("cuda:0", 30), ("cpu", 30)),
+        auto_vram_balance=False,
+    )
+    eng.setup(m, force_copy_lead=True)
+    configure_pipeline(eng, microbatches=2)
+    assert eng.pipeline is not None
+    out = eng.forward(x, t, context=c, **kw)
+    torch.testing.assert_close(out, ref, rtol=5e-3, atol=5e-3)
+    eng.release()
+
+
 def test_install_forward_cleanup_gpu():
     m = make_flux(dev="cuda:0", dtype=torch.bfloat16, tiny=True)
     eng = ParallelEngine(chain(("cuda:0", 100)), auto_vram_balance=False)
