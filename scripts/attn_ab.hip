@@ -4,9 +4,14 @@
 //
 // Variants:
 //   0: shipped v4
-//   1: + nt epilogue O stores + nt Q loads (keep K/V resident in the
-//        XCD's L2; O is written once, Q read once per block)
-//      + static s_setprio(1) for the younger wave half (guide T5 static)
+//   1: tr_b16 V path (guide T10): V stored ROW-major in LDS with a
+//      40-granule row stride (≡8 mod 32 -> conflict-free tr gather) and
+//      a (row&8)<<1 column XOR; written with plain b128 stores (the 16
+//      scalar V^T stores per thread disappear); PV A-fragments read with
+//      ds_read_b64_tr_b16 (2 per (chunk, tile), base VGPR + immediates).
+//      Gather semantics verified empirically (scripts/tr_probe.hip):
+//      within a 16-lane group, lane L reg j = element (L&3) of the
+//      granule addressed by lane (L>>2)+4j.
 //
 // Build: hipcc --offload-arch=gfx950 -O3 -std=c++17 attn_ab.hip -o attn_ab
 // Run:   ./attn_ab [rounds]
@@ -21,8 +26,10 @@
 
 using bf16 = __hip_bfloat16;
 typedef __attribute__((ext_vector_type(8))) short bf16x8;
+typedef __attribute__((ext_vector_type(4))) short s16x4;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 typedef __attribute__((ext_vector_type(16))) float f32x16;
+#define LDS_P __attribute__((address_space(3)))
 #define PA_DEV __device__ __forceinline__
 #define PA_LOG2E 1.4426950408889634f
 
@@ -46,11 +53,13 @@ __global__ __launch_bounds__(512, 2) void attn_ab_kernel(
     constexpr int WAVES = 8;
     constexpr int SUPER = 1;
     constexpr int DEPTH = 1;
-    constexpr bool NT_IO = (VAR == 1);
-    constexpr bool STATIC_PRIO = (VAR == 1);
+    constexpr bool NT_IO = false;
+    constexpr bool STATIC_PRIO = false;
+    constexpr bool TRV = (VAR == 1);   // tr_b16 V operand path
     constexpr int THREADS = WAVES * 64;
     constexpr int KPAD = D + 8;
     constexpr int VPAD = KVBLK + 8;
+    constexpr int VROW = 160;          // TRV row stride (elements): 40 granules
     constexpr int KK = D / 16;
     constexpr int NV = D / 32;
     constexpr int KVECS = (SUPER * KVBLK * D) / (8 * THREADS);
@@ -63,13 +72,17 @@ __global__ __launch_bounds__(512, 2) void attn_ab_kernel(
     constexpr bool VPAIR = false;
 
     __shared__ bf16 k_lds[SUPER * KVBLK * KPAD];
-    __shared__ bf16 v_lds[SUPER * D * VPAD];
+    __shared__ bf16 v_lds[TRV ? (KVBLK * VROW) : (SUPER * D * VPAD)];
 
     const int tid = threadIdx.x;
     const int lane = tid & 63;
     const int wid = tid >> 6;
     const int l32 = lane & 31;
     const int hi = lane >> 5;
+    // tr_b16 per-lane base: this lane supplies the granule
+    // (key = hi*8 + (p>>2) [+16c+4rd], col = 32n + 16*((G&1)^hi) + 4*(p&3))
+    const int trb = (hi * 8 + ((lane & 15) >> 2)) * VROW +
+                    16 * (((lane >> 4) & 1) ^ hi) + 4 * (lane & 3);
     if (STATIC_PRIO &&
         __builtin_amdgcn_readfirstlane(threadIdx.x) >= (WAVES / 2) * 64)
         __builtin_amdgcn_s_setprio(1);
@@ -152,7 +165,13 @@ __global__ __launch_bounds__(512, 2) void attn_ab_kernel(
             const int sub = row / KVBLK;          // 64-key sub-tile id
             const int srow = row % KVBLK;
             *reinterpret_cast<bf16x8*>(&k_lds[row * KPAD + col]) = kreg[slot][i];
-            if (VPAIR && D == 128) {
+            if (TRV) {
+                // row-major V, b128 store; column XOR by (row&8)<<1 keeps the
+                // tr gather conflict-free under either half-wave pairing
+                *reinterpret_cast<bf16x8*>(
+                    &v_lds[row * VROW + (col ^ ((row & 8) << 1))]) =
+                    vreg[slot][i];
+            } else if (VPAIR && D == 128) {
                 // pair keys (row, row^1) via shfl_xor(16): even-key threads
                 // write dims col..col+3 as b32, odd-key threads col+4..col+7
                 const bool even = ((lane >> 4) & 1) == 0;
@@ -276,10 +295,21 @@ __global__ __launch_bounds__(512, 2) void attn_ab_kernel(
         for (int c = 0; c < 4; ++c) {
 #pragma unroll
             for (int n = 0; n < NV; ++n) {
-                const int dim = n * 32 + l32;
-                const int gsw = ((2 * c + hi) ^ ((dim >> 3) & 7)) << 3;
-                bf16x8 va = *reinterpret_cast<const bf16x8*>(
-                    &v_lds[(sub * D + dim) * VPAD + gsw]);
+                bf16x8 va;
+                if (TRV) {
+                    s16x4 alo = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+                        (LDS_P s16x4*)&v_lds[trb + c * (16 * VROW) + n * 32]);
+                    s16x4 ahi = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+                        (LDS_P s16x4*)&v_lds[trb + c * (16 * VROW) +
+                                             4 * VROW + n * 32]);
+                    va = __builtin_shufflevector(alo, ahi,
+                                                 0, 1, 2, 3, 4, 5, 6, 7);
+                } else {
+                    const int dim = n * 32 + l32;
+                    const int gsw = ((2 * c + hi) ^ ((dim >> 3) & 7)) << 3;
+                    va = *reinterpret_cast<const bf16x8*>(
+                        &v_lds[(sub * D + dim) * VPAD + gsw]);
+                }
                 o_acc[n] = mfma32x32x16(va, pfrag[c], o_acc[n]);
             }
         }
