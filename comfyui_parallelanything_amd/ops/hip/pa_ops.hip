@@ -1120,9 +1120,21 @@ __global__ void pack_joint_qkv_kernel(
 // O^T = V^T · P^T (same P fragments serve as the B operand), keeping the
 // output row lane-local for the rescale and the 1/l epilogue.
 //
-// Geometry: 256-thread blocks (4 waves x 32 q-rows = 128 rows/block);
-// LDS = K[64][D+8] + swizzled V^T[D][72] = 35.8 KB -> two blocks co-resident
-// per CU, NOT barrier-synced against each other (phase overlap for free).
+// V operand via gfx950's ds_read_b64_tr_b16 hardware transpose-read (guide
+// T10): V is stored ROW-major ([key][dim], plain b128 stores straight from
+// the HBM staging registers — no per-element transpose pass) with a
+// 40-granule (160-element) row stride, which is ≡ 8 (mod 32) so each
+// tr-read cycle hits 32 distinct banks, plus a (key&8)<<1 column XOR that
+// keeps the two wave halves bank-disjoint. PV A-fragments are gathered by
+// two tr reads per (chunk, tile) from one base VGPR + immediate offsets.
+// Gather semantics verified empirically (scripts/tr_probe.hip): within a
+// 16-lane group, lane L reg j = element (L&3) of the granule addressed by
+// lane (L>>2)+4j. Measured +9.3% flux / +7.4% long-S over the swizzled
+// scalar-store image (scripts/attn_ab.hip, within-probe interleaved).
+//
+// Geometry: 512-thread blocks (8 waves x 32 q-rows = 256 rows/block);
+// LDS = K[64][D+8] + row-major V[64][160] = 37.9 KB (D=128) -> two blocks
+// co-resident per CU, NOT barrier-synced against each other.
 // ---------------------------------------------------------------------------
 typedef __attribute__((ext_vector_type(16))) float f32x16;
 
@@ -1151,19 +1163,23 @@ __global__ __launch_bounds__(512, 2) void attn_fwd_v4_kernel(
     constexpr int WAVES = 8;
     constexpr int THREADS = WAVES * 64;
     constexpr int KPAD = D + 8;
-    constexpr int VPAD = KVBLK + 8;
+    constexpr int VROW = 160;         // V row stride: 40 granules ≡ 8 mod 32
     constexpr int KK = D / 16;        // QK^T K-steps per 32-key tile
     constexpr int NV = D / 32;        // PV dim tiles
     constexpr int KVECS = (KVBLK * D) / (8 * THREADS);
 
     __shared__ bf16 k_lds[KVBLK * KPAD];
-    __shared__ bf16 v_lds[D * VPAD];
+    __shared__ bf16 v_lds[KVBLK * VROW];
 
     const int tid = threadIdx.x;
     const int lane = tid & 63;
     const int wid = tid >> 6;
     const int l32 = lane & 31;        // this lane's query row (within wave)
     const int hi = lane >> 5;         // half-wave id
+    // tr_b16 per-lane base: this lane supplies the granule
+    // (key = hi*8 + (p>>2) [+16c+4rd], col = 32n + 16*((G&1)^hi) + 4*(p&3))
+    const int trb = (hi * 8 + ((lane & 15) >> 2)) * VROW +
+                    16 * (((lane >> 4) & 1) ^ hi) + 4 * (lane & 3);
 
     long bh;
     int qtile;
@@ -1234,14 +1250,10 @@ __global__ __launch_bounds__(512, 2) void attn_fwd_v4_kernel(
             const int row = idx / (D / 8);
             const int col = (idx % (D / 8)) * 8;
             *reinterpret_cast<bf16x8*>(&k_lds[row * KPAD + col]) = kreg[i];
-#pragma unroll
-            for (int j = 0; j < 8; ++j) {
-                const int dim = col + j;
-                const int key_swz =
-                    (((row >> 3) ^ ((dim >> 3) & 7)) << 3) | (row & 7);
-                v_lds[dim * VPAD + key_swz] =
-                    __ushort_as_bfloat16((unsigned short)vreg[i][j]);
-            }
+            // row-major V, b128 store; column XOR by (row&8)<<1 keeps the
+            // tr gather conflict-free under either half-wave pairing
+            *reinterpret_cast<bf16x8*>(
+                &v_lds[row * VROW + (col ^ ((row & 8) << 1))]) = vreg[i];
         }
     };
 
@@ -1329,16 +1341,22 @@ __global__ __launch_bounds__(512, 2) void attn_fwd_v4_kernel(
         }
 
         // ---- PV: O^T[dim][row] += V^T chunk · P^T chunk ------------------
+        // A = V^T: lane holds V^T[n*32 + l32][16c + hi*8 + j], gathered
+        // from the row-major V image by ds_read_b64_tr_b16 (2 reads per
+        // (c, n): keys +0..3 and +4..7), base VGPR + immediate offsets.
         __builtin_amdgcn_s_setprio(1);
 #pragma unroll
         for (int c = 0; c < 4; ++c) {
 #pragma unroll
             for (int n = 0; n < NV; ++n) {
-                // A = V^T: lane holds V^T[n*32 + l32][16c + hi*8 + j]
-                const int dim = n * 32 + l32;
-                const int gsw = ((2 * c + hi) ^ ((dim >> 3) & 7)) << 3;
-                bf16x8 va = *reinterpret_cast<const bf16x8*>(
-                    &v_lds[dim * VPAD + gsw]);
+                short4v alo = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+                    (__attribute__((address_space(3))) short4v*)
+                        &v_lds[trb + c * (16 * VROW) + n * 32]);
+                short4v ahi = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+                    (__attribute__((address_space(3))) short4v*)
+                        &v_lds[trb + c * (16 * VROW) + 4 * VROW + n * 32]);
+                bf16x8 va = __builtin_shufflevector(alo, ahi,
+                                                    0, 1, 2, 3, 4, 5, 6, 7);
                 o_acc[n] = mfma32x32x16(va, pfrag[c], o_acc[n]);
             }
         }
