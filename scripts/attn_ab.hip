@@ -55,7 +55,9 @@ __global__ __launch_bounds__(512, 2) void attn_ab_kernel(
     constexpr int DEPTH = 1;
     constexpr bool NT_IO = false;
     constexpr bool STATIC_PRIO = false;
-    constexpr bool TRV = (VAR == 1);   // tr_b16 V operand path
+    constexpr bool TRV = true;         // tr_b16 V path (shipped baseline)
+    constexpr bool DEFER = (VAR == 1); // T13 defer-max RESCALE_THRESHOLD
+    constexpr float DEFER_THR = 8.0f;  // exp2 domain: P bounded by 2^8
     constexpr int THREADS = WAVES * 64;
     constexpr int KPAD = D + 8;
     constexpr int VPAD = KVBLK + 8;
@@ -250,7 +252,11 @@ __global__ __launch_bounds__(512, 2) void attn_ab_kernel(
                 mx = fmaxf(mx, sv);
             }
         mx = fmaxf(mx, __shfl_xor(mx, 32, 64));
-        const float mnew = fmaxf(m_run, mx);
+        // T13: if the max grew by < THR, keep the old max — alpha == 1
+        // exactly, so the O rescale below is skipped; P is then bounded by
+        // 2^THR instead of 1 (bf16-accum headroom covers it).
+        const float mnew = DEFER ? (mx <= m_run + DEFER_THR ? m_run : mx)
+                                 : fmaxf(m_run, mx);
         const float alpha = exp2x(m_run - mnew);
         m_run = mnew;
         float ps = 0.f;
