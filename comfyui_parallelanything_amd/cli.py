@@ -43,6 +43,9 @@ def main(argv=None):
                     help="pipeline-mode micro-batches (>1 overlaps stages)")
     ap.add_argument("--hip-graphs", action="store_true",
                     help="capture repeated same-shape forwards into hipGraphs")
+    ap.add_argument("--lora", default=None, metavar="PATH[:SCALE]",
+                    help="merge a LoRA .safetensors into the weights before "
+                         "replication (PEFT or kohya key conventions)")
     args = ap.parse_args(argv)
 
     devices = args.devices.split(",")
@@ -64,6 +67,13 @@ def main(argv=None):
     )
     make, make_inputs = MODELS[args.model]
     model = make(dev=chain.lead, dtype=dtype, tiny=tiny)
+
+    if args.lora:
+        from .models.lora import merge_lora_file
+
+        path, _, s = args.lora.partition(":")
+        n = merge_lora_file(model, path, scale=float(s) if s else 1.0)
+        print(f"[lora] merged {n} modules from {path}")
 
     engine = ParallelEngine(chain, workload_split=args.split,
                             auto_vram_balance=args.balance,
