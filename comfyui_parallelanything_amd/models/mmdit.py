@@ -58,6 +58,22 @@ class FluxConfig:
         return cls()
 
     @classmethod
+    def sd35_large(cls) -> "FluxConfig":
+        """SD3.5-Large-class MMDiT: joint (dual-stream) trunk only — no
+        single-stream tail — with 64-dim heads (exercises the D=64
+        attention path at MMDiT scale)."""
+        return cls(hidden=2432, num_heads=38, depth_double=38,
+                   depth_single=0, context_dim=4096, vec_dim=2048,
+                   axes_dim=(16, 24, 24), guidance_embed=False)
+
+    @classmethod
+    def sd35_tiny(cls) -> "FluxConfig":
+        return cls(in_channels=4, hidden=64, num_heads=4, depth_double=3,
+                   depth_single=0, context_dim=32, vec_dim=16,
+                   axes_dim=(4, 6, 6), time_embed_dim=32,
+                   guidance_embed=False)
+
+    @classmethod
     def tiny(cls) -> "FluxConfig":
         """CPU-test scale."""
         return cls(in_channels=4, hidden=64, num_heads=4, depth_double=2,

@@ -102,6 +102,27 @@ def sdxl_inputs(batch: int, px: int = 1024, dev="cpu", dtype=torch.bfloat16,
     return x, t, ctx, kw
 
 
+def make_sd3(dev="cpu", dtype=torch.bfloat16, tiny=False):
+    cfg = FluxConfig.sd35_tiny() if tiny else FluxConfig.sd35_large()
+    torch.manual_seed(0)
+    with torch.device(dev):
+        m = Flux(cfg)
+    return m.to(dtype=dtype).eval()
+
+
+def sd3_inputs(batch: int, px: int = 1024, dev="cpu", dtype=torch.bfloat16,
+               tiny=False, seed: int = 1234):
+    cfg = FluxConfig.sd35_tiny() if tiny else FluxConfig.sd35_large()
+    g = torch.Generator(device="cpu").manual_seed(seed)
+    hw = _latent_hw(px) if not tiny else 16
+    x = torch.randn(batch, cfg.in_channels, hw, hw, generator=g).to(dev, dtype)
+    t = torch.rand(batch, generator=g).to(dev, torch.float32)
+    txt_len = 154 if not tiny else 8
+    ctx = torch.randn(batch, txt_len, cfg.context_dim, generator=g).to(dev, dtype)
+    y = torch.randn(batch, cfg.vec_dim, generator=g).to(dev, dtype)
+    return x, t, ctx, {"y": y}
+
+
 def make_wan(dev="cpu", dtype=torch.bfloat16, tiny=False):
     cfg = WanConfig.tiny() if tiny else WanConfig.wan22_a14b()
     torch.manual_seed(0)
@@ -128,5 +149,6 @@ MODELS: Dict[str, Tuple[Callable, Callable]] = {
     "zimage": (make_zimage, zimage_inputs),
     "sd15": (make_sd15, sd15_inputs),
     "sdxl": (make_sdxl, sdxl_inputs),
+    "sd3": (make_sd3, sd3_inputs),
     "wan": (make_wan, wan_inputs),
 }

@@ -136,7 +136,7 @@ def test_vram_balancer_reads_hbm():
     assert free > 10_000, f"expected >10 GB free HBM, got {free} MB"
 
 
-@pytest.mark.parametrize("name", ["sdxl", "zimage", "wan"])
+@pytest.mark.parametrize("name", ["sdxl", "zimage", "sd3", "wan"])
 def test_model_families_gpu_tiny(name):
     from comfyui_parallelanything_amd.models.registry import MODELS
 
