@@ -196,6 +196,7 @@ def main():
                     "zimage": "Z-Image-class DiT 6B",
                     "sdxl": "SDXL-class UNet 2.6B",
                     "sd15": "SD1.5-class UNet",
+                    "sd3": "SD3.5-Large-class MMDiT 8B",
                     "wan": "WAN2.2-class video DiT 14B",
                 }[args.model] + (" [TINY DEBUG CONFIG]" if tiny else ""),
                 "global_batch": args.batch,
