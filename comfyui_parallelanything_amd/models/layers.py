@@ -38,6 +38,16 @@ class GELULinear(nn.Module):
         self.lin = nn.Linear(in_features, out_features, bias=True)
 
     def forward(self, x):
+        if not isinstance(self.lin, nn.Linear):
+            # fp8-swapped (models/quant.py): _scaled_mm has no GELU
+            # epilogue, so run the fp8 GEMM + the standalone gelu kernel —
+            # measured faster than keeping this GEMM bf16-fused.
+            y = self.lin(x)
+            if y.is_cuda:
+                from .. import ops
+
+                return ops.gelu_tanh(y)
+            return torch.nn.functional.gelu(y, approximate="tanh")
         if x.is_cuda:
             x2 = x.reshape(-1, x.shape[-1])
             out = torch._addmm_activation(
