@@ -52,13 +52,21 @@ class FP8Linear(nn.Module):
         self.out_dtype = out_dtype
         self.in_features = weight_fp8.shape[1]
         self.out_features = weight_fp8.shape[0]
-        # delayed activation scaling: running amax updated by the fused
-        # quant kernel each call; the scale used is the PREVIOUS call's.
-        # MUST live on the weight's device — the module is swapped in place
-        # and never .to()-ed afterwards.
+        # delayed activation scaling: running amax + next-call scale are
+        # maintained ENTIRELY by the quant kernel (slot 1 of a_amax is its
+        # block counter; see pa_ops.hip quant_fp8 fused epilogue) — zero
+        # per-call host-launched scale math. The scale used is the
+        # PREVIOUS call's. Buffers MUST live on the weight's device — the
+        # module is swapped in place and never .to()-ed afterwards.
         self.register_buffer(
             "a_amax",
-            torch.zeros(1, dtype=torch.float32, device=weight_fp8.device),
+            torch.zeros(2, dtype=torch.float32, device=weight_fp8.device),
+            persistent=False,
+        )
+        self.register_buffer(
+            "x_scale",
+            torch.full((1,), 1.0, dtype=torch.float32,
+                       device=weight_fp8.device),
             persistent=False,
         )
         self._warm = False
@@ -82,11 +90,14 @@ class FP8Linear(nn.Module):
                 and ops.hip_available("quant_fp8")):
             if not self._warm:
                 # first call: measure directly (still async, device-side)
-                self.a_amax.copy_(x2.abs().amax().reshape(1).float())
+                amax0 = x2.abs().amax().reshape(1).float()
+                self.a_amax[:1].copy_(amax0)
+                self.x_scale.copy_((amax0 / FP8_MAX).clamp(min=1e-12))
                 self._warm = True
-            x_scale = (self.a_amax / FP8_MAX).clamp(min=1e-12)
-            self.a_amax.mul_(0.999)  # slow decay lets the scale shrink
-            x8 = ops.quant_fp8(x2, x_scale, self.a_amax)
+            x_scale = self.x_scale
+            # the kernel quantizes with x_scale, updates the running amax,
+            # and writes the NEXT call's x_scale — no host-side scale math
+            x8 = ops.quant_fp8(x2, self.x_scale, self.a_amax)
         else:
             x_scale = (x2.abs().amax().float() / FP8_MAX).clamp(min=1e-12)
             x8 = (x2.float() / x_scale).clamp(-FP8_MAX, FP8_MAX).to(FP8)

@@ -493,9 +493,9 @@ __global__ void qk_norm_rope_kernel(bf16* __restrict__ q, bf16* __restrict__ k,
 // host syncs — the torch-level dynamic-quant path measured 7-12x this cost).
 __global__ void quant_fp8_bf16_kernel(const bf16* __restrict__ x,
                                       unsigned char* __restrict__ out,
-                                      const float* __restrict__ scale,
+                                      float* __restrict__ scale,
                                       float* __restrict__ amax_buf,
-                                      long total8) {
+                                      long total8, int fuse_scale) {
     const long stride = (long)gridDim.x * blockDim.x;
     const short8* xv = reinterpret_cast<const short8*>(x);
     const float inv_s = 1.0f / scale[0];
@@ -532,6 +532,24 @@ __global__ void quant_fp8_bf16_kernel(const bf16* __restrict__ x,
             m = fmaxf(m, scratch[i]);
         atomicMax(reinterpret_cast<unsigned int*>(amax_buf),
                   __float_as_uint(m));
+        // fused delayed-scaling epilogue (amax_buf[1] = block counter):
+        // the LAST block to finish decays the running amax and writes the
+        // NEXT call's scale — replaces two host-launched micro-kernels per
+        // Linear call. Every other block loaded scale[0] at entry (before
+        // its own counter increment), so the update cannot race a reader.
+        if (fuse_scale) {
+            __threadfence();
+            const unsigned int done = atomicAdd(
+                reinterpret_cast<unsigned int*>(&amax_buf[1]), 1u);
+            if (done == gridDim.x - 1) {
+                __threadfence();
+                reinterpret_cast<unsigned int*>(amax_buf)[1] = 0u;
+                const float next = amax_buf[0] * 0.999f;  // slow decay
+                amax_buf[0] = next;
+                scale[0] = fmaxf(next / 448.f, 1e-12f);
+                __threadfence();
+            }
+        }
     }
 }
 
@@ -1495,11 +1513,14 @@ at::Tensor quant_fp8(at::Tensor x, at::Tensor scale, at::Tensor amax_buf) {
     auto out = at::empty(xc.sizes(), xc.options().dtype(at::kFloat8_e4m3fn));
     const long total8 = xc.numel() / 8;
     const int blocks = (int)std::min<long>((total8 + 255) / 256, 4096);
+    // amax_buf with a second slot (counter) opts into the fused
+    // delayed-scaling epilogue: the kernel itself writes the next scale.
+    const int fuse_scale = amax_buf.numel() >= 2 ? 1 : 0;
     hipLaunchKernelGGL(quant_fp8_bf16_kernel, dim3(blocks), dim3(256), 0,
                        cur_stream(), (const bf16*)xc.data_ptr(),
                        (unsigned char*)out.data_ptr(),
                        scale.data_ptr<float>(), amax_buf.data_ptr<float>(),
-                       total8);
+                       total8, fuse_scale);
     return out;
 }
 

@@ -69,3 +69,30 @@ def test_fp8_linear_delayed_scaling_stable():
     for out in outs[1:]:
         rel = (out - ref).norm() / ref.norm()
         assert rel < 0.06, f"fp8 delayed-scale error {rel:.4f}"
+    # the kernel-maintained scale tracks amax/448 (decayed) and the block
+    # counter (slot 1) is reset after every call
+    amax = q.a_amax[0].item()
+    assert q.a_amax[1].item() == 0.0
+    assert abs(q.x_scale.item() - max(amax / 448.0, 1e-12)) < 1e-6
+
+
+def test_quant_fp8_fused_scale_epilogue():
+    from comfyui_parallelanything_amd import ops
+
+    if not ops.hip_available("quant_fp8"):
+        pytest.skip("no quant_fp8 in extension")
+    torch.manual_seed(2)
+    x = torch.randn(512, 2048, device="cuda", dtype=torch.bfloat16) * 2
+    true_amax = x.abs().amax().item()
+    scale = torch.tensor([true_amax / 448.0], device="cuda")
+    amax = torch.zeros(2, device="cuda")  # slot 1 = kernel block counter
+    x8 = ops.quant_fp8(x, scale, amax)
+    ref = (x.float() / (true_amax / 448.0)).clamp(-448, 448).to(
+        torch.float8_e4m3fn
+    )
+    torch.testing.assert_close(x8.float(), ref.float(), rtol=0, atol=0)
+    # epilogue: amax decayed once, scale rewritten for the next call,
+    # counter back to zero
+    assert abs(amax[0].item() - true_amax * 0.999) < 1e-3
+    assert amax[1].item() == 0.0
+    assert abs(scale.item() - true_amax * 0.999 / 448.0) < 1e-6
