@@ -90,7 +90,13 @@ def test_quant_fp8_fused_scale_epilogue():
     ref = (x.float() / (true_amax / 448.0)).clamp(-448, 448).to(
         torch.float8_e4m3fn
     )
-    torch.testing.assert_close(x8.float(), ref.float(), rtol=0, atol=0)
+    # kernel multiplies by the reciprocal; at fp8 rounding boundaries that
+    # can land one quantum off the divide — compare within one ULP of fp8
+    diff = (x8.float() - ref.float()).abs()
+    ulp = torch.maximum(ref.float().abs() * 2 ** -3,
+                        torch.full_like(diff, 2 ** -9))
+    assert (diff <= ulp).all(), f"max diff {diff.max().item()}"
+    assert (diff > 0).float().mean().item() < 0.001  # almost all exact
     # epilogue: amax decayed once, scale rewritten for the next call,
     # counter back to zero
     assert abs(amax[0].item() - true_amax * 0.999) < 1e-3
