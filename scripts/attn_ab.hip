@@ -364,10 +364,11 @@ __global__ __launch_bounds__(512, 2) void attn_ab_kernel(
 
 int main(int argc, char** argv) {
     const int rounds = argc > 1 ? atoi(argv[1]) : 8;
-    struct Shape { int B, H, S; const char* name; };
-    Shape shapes[2] = {{8, 24, 4608, "flux"}, {1, 8, 30720, "long"}};
+    struct Shape { int B, H, S, D; const char* name; };
+    Shape shapes[3] = {{8, 24, 4608, 128, "flux"}, {1, 8, 30720, 128, "long"},
+                       {8, 38, 4250, 64, "sd3-d64"}};
     for (auto& sh : shapes) {
-        const int B = sh.B, H = sh.H, S = sh.S, D = 128;
+        const int B = sh.B, H = sh.H, S = sh.S, D = sh.D;
         const long n = (long)B * S * H * D;
         bf16 *q, *k, *v, *o;
         HIP_CHECK(hipMalloc(&q, n * 2));
@@ -394,7 +395,14 @@ int main(int argc, char** argv) {
         const double tf = 4.0 * B * H * (double)S * S * D / 1e12;
         auto run = [&](int var) {
             dim3 grid(((S + 255) / 256) * B * H), blk(512);
-            if (var == 0)
+            if (D == 64) {
+                if (var == 0)
+                    hipLaunchKernelGGL((attn_ab_kernel<64, 0>), grid, blk, 0,
+                                       0, q, k, v, o, S, S, scale, H);
+                else
+                    hipLaunchKernelGGL((attn_ab_kernel<64, 1>), grid, blk, 0,
+                                       0, q, k, v, o, S, S, scale, H);
+            } else if (var == 0)
                 hipLaunchKernelGGL((attn_ab_kernel<128, 0>), grid, blk, 0, 0,
                                    q, k, v, o, S, S, scale, H);
             else
