@@ -15,6 +15,7 @@ Run:  python -m comfyui_parallelanything_amd.serve \
 # NOTE: no `from __future__ import annotations` here — it stringifies the
 # request-model annotations defined inside create_app(), and FastAPI then
 # cannot resolve them (the body model silently degrades to a query param).
+import threading
 import time
 from typing import List, Optional
 
@@ -66,6 +67,9 @@ def create_app(
     app = FastAPI(title="parallelanything-amd")
     app.state.engine = engine
     app.state.model = model
+    # sync endpoints run in Starlette's threadpool: serialize engine use
+    # (the per-device stream scheduler is single-caller by design)
+    gen_lock = threading.Lock()
 
     @app.get("/healthz")
     def healthz():
@@ -84,19 +88,20 @@ def create_app(
                                      f"have {sorted(SAMPLERS)}")
         if not 1 <= req.batch <= 64 or not 1 <= req.steps <= 200:
             raise HTTPException(400, "batch must be 1..64, steps 1..200")
-        torch.manual_seed(req.seed)
-        x, _, c, kw = make_inputs(req.batch, dev=chain.lead, dtype=dtype,
-                                  tiny=tiny)
-        sig = (karras_sigmas(req.steps) if model_name in ("sd15", "sdxl")
-               else flow_sigmas(req.steps, shift=req.shift))
-        timer.start()
-        t0 = time.perf_counter()
-        with torch.no_grad():
-            out = SAMPLERS[req.sampler](model, x, sig, context=c, **kw)
-        if torch.cuda.is_available():
-            torch.cuda.synchronize()
-        dt = time.perf_counter() - t0
-        timer.stop(images=req.batch)
+        with gen_lock:
+            torch.manual_seed(req.seed)
+            x, _, c, kw = make_inputs(req.batch, dev=chain.lead, dtype=dtype,
+                                      tiny=tiny)
+            sig = (karras_sigmas(req.steps) if model_name in ("sd15", "sdxl")
+                   else flow_sigmas(req.steps, shift=req.shift))
+            timer.start()
+            t0 = time.perf_counter()
+            with torch.no_grad():
+                out = SAMPLERS[req.sampler](model, x, sig, context=c, **kw)
+            if torch.cuda.is_available():
+                torch.cuda.synchronize()
+            dt = time.perf_counter() - t0
+            timer.stop(images=req.batch)
         of = out.float()
         return {
             "shape": list(out.shape),
