@@ -15,6 +15,14 @@ replays the full forward, and each wrapped block either
 After the last block, its owner sends the hidden state to rank 0, whose
 final layers produce the real output; other ranks' outputs are dummies.
 Works over gloo for the CPU tests — the call pattern is identical to RCCL.
+
+Micro-batching note: the in-process pipeline supports GPipe-style
+micro-batching (pipeline.py, batches too small for DP). Here it is
+deliberately absent: in process-group mode any batch > 1 routes to DP,
+which beats pipelining outright (no bubble), and batch == 1 has nothing
+to split. Overlapping micro-batches across ranks would also require
+tagged p2p matching, which RCCL does not support (NCCL/RCCL send/recv
+ignores tags) — concurrent per-thread send/recv streams would mismatch.
 """
 from __future__ import annotations
 
