@@ -144,6 +144,26 @@ def test_release_clears_state(sd15):
     assert not eng.replicas and not eng.streams and eng.pipeline is None
 
 
+def test_golden_three_device_asymmetric(sd15):
+    """70/20/10 [cpu,cpu,cpu]: min-1 floors + remainder; == single device."""
+    x, t, c, kw = sd15_inputs(10, tiny=True)
+    ref = sd15(x, t, context=c, **kw)
+    eng = ParallelEngine(cpu_chain(70, 20, 10), auto_vram_balance=False)
+    eng.setup(sd15)
+    out = eng.forward(x, t, context=c, **kw)
+    torch.testing.assert_close(out, ref, rtol=1e-4, atol=1e-5)
+
+
+def test_golden_more_devices_than_batch_dp_threshold(sd15):
+    """batch == n_devices: every device gets exactly one sample."""
+    x, t, c, kw = sd15_inputs(3, tiny=True)
+    ref = sd15(x, t, context=c, **kw)
+    eng = ParallelEngine(cpu_chain(34, 33, 33), auto_vram_balance=False)
+    eng.setup(sd15)
+    out = eng.forward(x, t, context=c, **kw)
+    torch.testing.assert_close(out, ref, rtol=1e-4, atol=1e-5)
+
+
 def test_hip_graphs_cpu_falls_back_to_eager(sd15):
     """use_hip_graphs on a CPU chain must be a transparent no-op."""
     x, t, c, kw = sd15_inputs(2, tiny=True)
