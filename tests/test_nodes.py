@@ -61,6 +61,9 @@ def test_anything_schema_defaults():
     assert opts["auto_vram_balance"][1]["default"] is True
     assert opts["purge_cache"][1]["default"] is True
     assert opts["purge_models"][1]["default"] is False
+    # extensions default OFF: stock node behavior == reference behavior
+    assert opts["pipeline_microbatches"][1]["default"] == 1
+    assert opts["use_hip_graphs"][1]["default"] is False
 
 
 def test_chain_building_via_nodes():
