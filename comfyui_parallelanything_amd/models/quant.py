@@ -69,6 +69,18 @@ class FP8Linear(nn.Module):
                        device=weight_fp8.device),
             persistent=False,
         )
+        # Snapshot of the scale the quant kernel actually used this call:
+        # its epilogue overwrites x_scale[0] with the NEXT call's scale, so
+        # _scaled_mm must dequantize from this buffer, never x_scale itself
+        # (same-buffer use multiplied the output by next_scale/used_scale).
+        # A persistent separate buffer keeps all kernel pointers fixed
+        # across calls, so hipGraph capture/replay stays valid.
+        self.register_buffer(
+            "x_scale_used",
+            torch.full((1,), 1.0, dtype=torch.float32,
+                       device=weight_fp8.device),
+            persistent=False,
+        )
         self._warm = False
 
     @classmethod
@@ -94,10 +106,12 @@ class FP8Linear(nn.Module):
                 self.a_amax[:1].copy_(amax0)
                 self.x_scale.copy_((amax0 / FP8_MAX).clamp(min=1e-12))
                 self._warm = True
-            x_scale = self.x_scale
             # the kernel quantizes with x_scale, updates the running amax,
-            # and writes the NEXT call's x_scale — no host-side scale math
-            x8 = ops.quant_fp8(x2, self.x_scale, self.a_amax)
+            # writes the NEXT call's x_scale, and snapshots the scale it
+            # used into x_scale_used — no host-side scale math
+            x8 = ops.quant_fp8(x2, self.x_scale, self.a_amax,
+                               scale_used=self.x_scale_used)
+            x_scale = self.x_scale_used
         else:
             x_scale = (x2.abs().amax().float() / FP8_MAX).clamp(min=1e-12)
             x8 = (x2.float() / x_scale).clamp(-FP8_MAX, FP8_MAX).to(FP8)

@@ -269,11 +269,17 @@ def pack_joint_qkv(txt_qkv, img_qkv, wq_t, wk_t, wq_i, wk_i, cs,
 
 
 def quant_fp8(x: torch.Tensor, scale: torch.Tensor,
-              amax_buf: torch.Tensor) -> torch.Tensor:
+              amax_buf: torch.Tensor,
+              scale_used: torch.Tensor | None = None) -> torch.Tensor:
     """Fused bf16 -> e4m3fn cast with running-amax update (delayed scaling).
+
+    The fused epilogue overwrites ``scale[0]`` with the NEXT call's scale;
+    pass a separate ``scale_used`` buffer to receive a snapshot of the scale
+    this call actually quantized with (what _scaled_mm must dequantize by).
     GPU-only; used by the fp8 serving mode."""
     ext = _require_ext("quant_fp8")
-    return ext.quant_fp8(x, scale, amax_buf)
+    return ext.quant_fp8(x, scale, amax_buf,
+                         scale if scale_used is None else scale_used)
 
 
 def gelu_tanh(x: torch.Tensor) -> torch.Tensor:

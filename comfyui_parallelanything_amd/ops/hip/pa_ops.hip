@@ -421,14 +421,19 @@ __global__ void qk_norm_rope_kernel(bf16* __restrict__ q, bf16* __restrict__ k,
     const int lane = threadIdx.x & 63;
     const long wave = ((long)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
     const int pairs = D / 2;
-    if (lane >= pairs) return;
+    // D < 128 leaves lanes >= pairs with no data, but they MUST stay alive
+    // through the 64-lane __shfl_xor reductions below (an early return here
+    // would feed undefined registers into the cross-lane sum): inactive
+    // lanes load from a clamped address, contribute 0, and skip the store.
+    const bool act = lane < pairs;
+    const int ln = act ? lane : 0;
 
     unsigned int uq[RPW], uk[RPW];
     unsigned int* qp[RPW];
     unsigned int* kp[RPW];
     float cvec[RPW], svec[RPW];
-    const unsigned int uwq = reinterpret_cast<const unsigned int*>(wq)[lane];
-    const unsigned int uwk = reinterpret_cast<const unsigned int*>(wk)[lane];
+    const unsigned int uwq = reinterpret_cast<const unsigned int*>(wq)[ln];
+    const unsigned int uwk = reinterpret_cast<const unsigned int*>(wk)[ln];
 #pragma unroll
     for (int i = 0; i < RPW; ++i) {
         const long row = wave * RPW + i;
@@ -441,10 +446,10 @@ __global__ void qk_norm_rope_kernel(bf16* __restrict__ q, bf16* __restrict__ k,
             q + b * q_bs + (long)sj * q_ss + (long)h * q_hs);
         kp[i] = reinterpret_cast<unsigned int*>(
             k + b * k_bs + (long)sj * k_ss + (long)h * k_hs);
-        uq[i] = qp[i][lane];
-        uk[i] = kp[i][lane];
-        cvec[i] = cs[((long)sj * pairs + lane) * 2 + 0];
-        svec[i] = cs[((long)sj * pairs + lane) * 2 + 1];
+        uq[i] = qp[i][ln];
+        uk[i] = kp[i][ln];
+        cvec[i] = cs[((long)sj * pairs + ln) * 2 + 0];
+        svec[i] = cs[((long)sj * pairs + ln) * 2 + 1];
     }
     const float wq0 = bf2f(__ushort_as_bfloat16((unsigned short)(uwq & 0xffff)));
     const float wq1 = bf2f(__ushort_as_bfloat16((unsigned short)(uwq >> 16)));
@@ -456,7 +461,7 @@ __global__ void qk_norm_rope_kernel(bf16* __restrict__ q, bf16* __restrict__ k,
         {
             float a0 = bf2f(__ushort_as_bfloat16((unsigned short)(uq[i] & 0xffff)));
             float a1 = bf2f(__ushort_as_bfloat16((unsigned short)(uq[i] >> 16)));
-            float ss_ = a0 * a0 + a1 * a1;
+            float ss_ = act ? a0 * a0 + a1 * a1 : 0.f;
 #pragma unroll
             for (int off = 32; off > 0; off >>= 1)
                 ss_ += __shfl_xor(ss_, off, 64);
@@ -465,13 +470,14 @@ __global__ void qk_norm_rope_kernel(bf16* __restrict__ q, bf16* __restrict__ k,
             a1 = a1 * rr * wq1;
             const float o0 = a0 * cvec[i] - a1 * svec[i];
             const float o1 = a0 * svec[i] + a1 * cvec[i];
-            qp[i][lane] = (unsigned int)__bfloat16_as_ushort(f2bf(o0)) |
-                          ((unsigned int)__bfloat16_as_ushort(f2bf(o1)) << 16);
+            if (act)
+                qp[i][lane] = (unsigned int)__bfloat16_as_ushort(f2bf(o0)) |
+                              ((unsigned int)__bfloat16_as_ushort(f2bf(o1)) << 16);
         }
         {
             float a0 = bf2f(__ushort_as_bfloat16((unsigned short)(uk[i] & 0xffff)));
             float a1 = bf2f(__ushort_as_bfloat16((unsigned short)(uk[i] >> 16)));
-            float ss_ = a0 * a0 + a1 * a1;
+            float ss_ = act ? a0 * a0 + a1 * a1 : 0.f;
 #pragma unroll
             for (int off = 32; off > 0; off >>= 1)
                 ss_ += __shfl_xor(ss_, off, 64);
@@ -480,8 +486,9 @@ __global__ void qk_norm_rope_kernel(bf16* __restrict__ q, bf16* __restrict__ k,
             a1 = a1 * rr * wk1;
             const float o0 = a0 * cvec[i] - a1 * svec[i];
             const float o1 = a0 * svec[i] + a1 * cvec[i];
-            kp[i][lane] = (unsigned int)__bfloat16_as_ushort(f2bf(o0)) |
-                          ((unsigned int)__bfloat16_as_ushort(f2bf(o1)) << 16);
+            if (act)
+                kp[i][lane] = (unsigned int)__bfloat16_as_ushort(f2bf(o0)) |
+                              ((unsigned int)__bfloat16_as_ushort(f2bf(o1)) << 16);
         }
     }
 }
@@ -495,10 +502,12 @@ __global__ void quant_fp8_bf16_kernel(const bf16* __restrict__ x,
                                       unsigned char* __restrict__ out,
                                       float* __restrict__ scale,
                                       float* __restrict__ amax_buf,
+                                      float* scale_used,  // may alias scale
                                       long total8, int fuse_scale) {
     const long stride = (long)gridDim.x * blockDim.x;
     const short8* xv = reinterpret_cast<const short8*>(x);
-    const float inv_s = 1.0f / scale[0];
+    const float s_entry = scale[0];
+    const float inv_s = 1.0f / s_entry;
     float local_amax = 0.f;
     for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total8;
          i += stride) {
@@ -546,6 +555,13 @@ __global__ void quant_fp8_bf16_kernel(const bf16* __restrict__ x,
                 reinterpret_cast<unsigned int*>(amax_buf)[1] = 0u;
                 const float next = amax_buf[0] * 0.999f;  // slow decay
                 amax_buf[0] = next;
+                // Snapshot the scale actually USED for this quantization
+                // BEFORE overwriting scale[0] with the next call's value —
+                // _scaled_mm must dequantize with scale_used, not next
+                // (the two differ whenever the activation amax moves).
+                // scale_used may alias scale (legacy callers): skip the
+                // snapshot then, preserving the old in-place behavior.
+                if (scale_used != scale) scale_used[0] = s_entry;
                 scale[0] = fmaxf(next / 448.f, 1e-12f);
                 __threadfence();
             }
@@ -1065,7 +1081,10 @@ __global__ void pack_joint_qkv_kernel(
     const long wave = ((long)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
     const int S = T + Si;
     const int pairs = D / 2;
-    if (lane >= pairs) return;
+    // Same D<128 rule as qk_norm_rope_kernel: lanes >= pairs stay alive for
+    // the 64-lane reductions (contributing 0), load clamped, never store.
+    const bool act = lane < pairs;
+    const int ln = act ? lane : 0;
 
     unsigned int uq[RPW], uk[RPW], uv[RPW], uwq[RPW], uwk[RPW];
     long obase[RPW];
@@ -1083,13 +1102,13 @@ __global__ void pack_joint_qkv_kernel(
             ? txt + b * t_bs + (long)sj * t_ss + (long)h * t_hs
             : img + b * i_bs + (long)(sj - T) * i_ss + (long)h * i_hs;
         const long qs = is_txt ? t_qs : i_qs;
-        uq[i] = reinterpret_cast<const unsigned int*>(src)[lane];
-        uk[i] = reinterpret_cast<const unsigned int*>(src + qs)[lane];
-        uv[i] = reinterpret_cast<const unsigned int*>(src + 2 * qs)[lane];
-        uwq[i] = reinterpret_cast<const unsigned int*>(is_txt ? wq_t : wq_i)[lane];
-        uwk[i] = reinterpret_cast<const unsigned int*>(is_txt ? wk_t : wk_i)[lane];
-        cvec[i] = cs[((long)sj * pairs + lane) * 2 + 0];
-        svec[i] = cs[((long)sj * pairs + lane) * 2 + 1];
+        uq[i] = reinterpret_cast<const unsigned int*>(src)[ln];
+        uk[i] = reinterpret_cast<const unsigned int*>(src + qs)[ln];
+        uv[i] = reinterpret_cast<const unsigned int*>(src + 2 * qs)[ln];
+        uwq[i] = reinterpret_cast<const unsigned int*>(is_txt ? wq_t : wq_i)[ln];
+        uwk[i] = reinterpret_cast<const unsigned int*>(is_txt ? wk_t : wk_i)[ln];
+        cvec[i] = cs[((long)sj * pairs + ln) * 2 + 0];
+        svec[i] = cs[((long)sj * pairs + ln) * 2 + 1];
         obase[i] = (((long)b * S + sj) * H + h) * (D / 2);
     }
 #pragma unroll
@@ -1099,30 +1118,32 @@ __global__ void pack_joint_qkv_kernel(
         {   // q: rms + weight + rope
             float a0 = bf2f(__ushort_as_bfloat16((unsigned short)(uq[i] & 0xffff)));
             float a1 = bf2f(__ushort_as_bfloat16((unsigned short)(uq[i] >> 16)));
-            float ss_ = a0 * a0 + a1 * a1;
+            float ss_ = act ? a0 * a0 + a1 * a1 : 0.f;
 #pragma unroll
             for (int off = 32; off > 0; off >>= 1) ss_ += __shfl_xor(ss_, off, 64);
             const float rr = rsqrtf(ss_ / (float)D + eps);
             a0 = a0 * rr * bf2f(__ushort_as_bfloat16((unsigned short)(uwq[i] & 0xffff)));
             a1 = a1 * rr * bf2f(__ushort_as_bfloat16((unsigned short)(uwq[i] >> 16)));
-            reinterpret_cast<unsigned int*>(oq)[obase[i] + lane] =
-                (unsigned int)__bfloat16_as_ushort(f2bf(a0 * c - a1 * sn)) |
-                ((unsigned int)__bfloat16_as_ushort(f2bf(a0 * sn + a1 * c)) << 16);
+            if (act)
+                reinterpret_cast<unsigned int*>(oq)[obase[i] + lane] =
+                    (unsigned int)__bfloat16_as_ushort(f2bf(a0 * c - a1 * sn)) |
+                    ((unsigned int)__bfloat16_as_ushort(f2bf(a0 * sn + a1 * c)) << 16);
         }
         {   // k
             float a0 = bf2f(__ushort_as_bfloat16((unsigned short)(uk[i] & 0xffff)));
             float a1 = bf2f(__ushort_as_bfloat16((unsigned short)(uk[i] >> 16)));
-            float ss_ = a0 * a0 + a1 * a1;
+            float ss_ = act ? a0 * a0 + a1 * a1 : 0.f;
 #pragma unroll
             for (int off = 32; off > 0; off >>= 1) ss_ += __shfl_xor(ss_, off, 64);
             const float rr = rsqrtf(ss_ / (float)D + eps);
             a0 = a0 * rr * bf2f(__ushort_as_bfloat16((unsigned short)(uwk[i] & 0xffff)));
             a1 = a1 * rr * bf2f(__ushort_as_bfloat16((unsigned short)(uwk[i] >> 16)));
-            reinterpret_cast<unsigned int*>(ok)[obase[i] + lane] =
-                (unsigned int)__bfloat16_as_ushort(f2bf(a0 * c - a1 * sn)) |
-                ((unsigned int)__bfloat16_as_ushort(f2bf(a0 * sn + a1 * c)) << 16);
+            if (act)
+                reinterpret_cast<unsigned int*>(ok)[obase[i] + lane] =
+                    (unsigned int)__bfloat16_as_ushort(f2bf(a0 * c - a1 * sn)) |
+                    ((unsigned int)__bfloat16_as_ushort(f2bf(a0 * sn + a1 * c)) << 16);
         }
-        reinterpret_cast<unsigned int*>(ov)[obase[i] + lane] = uv[i];
+        if (act) reinterpret_cast<unsigned int*>(ov)[obase[i] + lane] = uv[i];
     }
 }
 
@@ -1505,7 +1526,8 @@ std::vector<at::Tensor> pack_joint_qkv(at::Tensor txt_qkv, at::Tensor img_qkv,
 }
 
 
-at::Tensor quant_fp8(at::Tensor x, at::Tensor scale, at::Tensor amax_buf) {
+at::Tensor quant_fp8(at::Tensor x, at::Tensor scale, at::Tensor amax_buf,
+                     at::Tensor scale_used) {
     CHECK_GPU(x);
     TORCH_CHECK(x.scalar_type() == at::kBFloat16, "quant_fp8: bf16 input");
     auto xc = x.contiguous();
@@ -1514,13 +1536,15 @@ at::Tensor quant_fp8(at::Tensor x, at::Tensor scale, at::Tensor amax_buf) {
     const long total8 = xc.numel() / 8;
     const int blocks = (int)std::min<long>((total8 + 255) / 256, 4096);
     // amax_buf with a second slot (counter) opts into the fused
-    // delayed-scaling epilogue: the kernel itself writes the next scale.
+    // delayed-scaling epilogue: the kernel itself writes the next scale
+    // into scale[0] and snapshots the scale it quantized with into
+    // scale_used[0] (pass scale_used=scale to skip the snapshot).
     const int fuse_scale = amax_buf.numel() >= 2 ? 1 : 0;
     hipLaunchKernelGGL(quant_fp8_bf16_kernel, dim3(blocks), dim3(256), 0,
                        cur_stream(), (const bf16*)xc.data_ptr(),
                        (unsigned char*)out.data_ptr(),
                        scale.data_ptr<float>(), amax_buf.data_ptr<float>(),
-                       total8, fuse_scale);
+                       scale_used.data_ptr<float>(), total8, fuse_scale);
     return out;
 }
 

@@ -224,26 +224,30 @@ class ParallelEngine:
     def _harvest_times(self):
         if not self._pending_times:
             return
-        done = []
+        still_pending = []
         for dev, size, rec in self._pending_times:
-            if isinstance(rec, float):
+            if isinstance(rec, float):  # cpu worker: wall-clock sample
                 self.balancer.record(dev, size, rec)
-                done.append((dev, size, rec))
-            else:
-                ev0, ev1 = rec
-                try:
-                    if ev1.query():
-                        self.balancer.record(
-                            dev, size, ev0.elapsed_time(ev1) / 1000.0
-                        )
-                        done.append((dev, size, rec))
-                except Exception:  # noqa: BLE001
-                    done.append((dev, size, rec))
-        self._pending_times = [p for p in self._pending_times if p not in done]
+                continue
+            ev0, ev1 = rec
+            try:
+                if not ev1.query():
+                    still_pending.append((dev, size, rec))
+                    continue
+                self.balancer.record(dev, size, ev0.elapsed_time(ev1) / 1000.0)
+            except Exception as err:  # noqa: BLE001
+                # drop the sample but say so — this feeds the load balancer
+                log.debug("balancer timing sample dropped on %s: %r", dev, err)
+        self._pending_times = still_pending
 
     def _data_parallel_impl(self, devices, sizes, batch, x, timesteps, context, kwargs):
         if self.balancer is None and self.auto_vram_balance:
-            self.balancer = AdaptiveBalancer(devices, list(self.chain.weights))
+            # Construct from the FULL chain, not this step's active subset:
+            # a size-0 device dropped on the first step would otherwise
+            # pair user weights with the wrong devices in weights().
+            self.balancer = AdaptiveBalancer(
+                list(self.chain.devices), list(self.chain.weights)
+            )
         x_chunks = split_batch(x, sizes)
         t_chunks = split_batch(timesteps, sizes)
         c_chunks = split_batch(context, sizes) if context is not None else None
@@ -261,8 +265,11 @@ class ParallelEngine:
         done_events: List[Optional[torch.cuda.Event]] = [None] * len(devices)
         keep_alive: List[Any] = []  # pin async intermediates until gathered
 
-        # Launch phase: one host thread, one stream per GPU.
-        for i, dev in enumerate(devices):
+        # Launch phase: one host thread, one stream per GPU. GPU enqueues
+        # go FIRST — a synchronous cpu worker early in a hybrid chain must
+        # not delay later stream enqueues (it computes while they run).
+        for i in self._launch_order(devices):
+            dev = devices[i]
             stream = self.streams.get(dev)
             try:
                 if stream is None:  # cpu worker: synchronous
@@ -343,6 +350,15 @@ class ParallelEngine:
         out = concatenate_results(results, dim=0)
         del keep_alive
         return out
+
+    def _launch_order(self, devices) -> List[int]:
+        """Indices with GPU (stream-backed) devices first, cpu workers last,
+        relative order preserved within each class. The cpu worker runs
+        synchronously on the launcher thread; putting it last means every
+        GPU stream is already busy while it computes."""
+        gpu = [i for i, d in enumerate(devices) if self.streams.get(d) is not None]
+        cpu = [i for i, d in enumerate(devices) if self.streams.get(d) is None]
+        return gpu + cpu
 
     @staticmethod
     @contextlib.contextmanager

@@ -36,7 +36,11 @@ def _sig(x: Any):
     if x is None:
         return None
     if isinstance(x, torch.Tensor):
-        return (tuple(x.shape), x.dtype)
+        # strides are part of the key: a same-shape NON-contiguous view
+        # (e.g. a transpose) has different semantics than the contiguous
+        # buffer a graph was captured on — replaying that graph after a
+        # plain copy_ would silently produce wrong output.
+        return (tuple(x.shape), tuple(x.stride()), x.dtype)
     return NotImplemented  # non-tensor: not graphable
 
 

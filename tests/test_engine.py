@@ -195,6 +195,35 @@ def test_graph_runner_key_rules():
     )
     # non-tensor kwarg -> not graphable
     assert r.key_for("cuda:0", x, t, None, {"flag": True}) is None
+    # same shape but different STRIDES (transposed view) -> different key:
+    # a graph captured on contiguous buffers must not replay on a view
+    # with different layout semantics
+    sq = torch.zeros(3, 3)
+    assert (
+        r.key_for("cuda:0", sq.t(), t, None, {})
+        != r.key_for("cuda:0", sq, t, None, {})
+    )
+
+
+def test_launch_order_gpu_before_cpu():
+    """Hybrid chains: stream-backed devices enqueue before cpu workers
+    (the cpu chunk computes while GPU streams are already busy)."""
+    from comfyui_parallelanything_amd.parallel.chain import DeviceChain
+    from comfyui_parallelanything_amd.parallel.engine import ParallelEngine
+
+    eng = ParallelEngine(
+        DeviceChain(
+            devices=("cpu", "cuda:0", "cuda:1"), weights=(0.34, 0.33, 0.33)
+        ),
+        auto_vram_balance=False,
+    )
+    # simulate stream-backed GPUs without hardware
+    eng.streams = {"cpu": None, "cuda:0": object(), "cuda:1": object()}
+    order = eng._launch_order(["cpu", "cuda:0", "cuda:1"])
+    assert order == [1, 2, 0]
+    # all-cpu chains keep natural order
+    eng.streams = {"cpu": None}
+    assert eng._launch_order(["cpu", "cpu"]) == [0, 1]
 
 
 def test_single_device_chain_routes_lead_only(sd15):

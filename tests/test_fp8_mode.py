@@ -102,3 +102,51 @@ def test_quant_fp8_fused_scale_epilogue():
     assert abs(amax[0].item() - true_amax * 0.999) < 1e-3
     assert amax[1].item() == 0.0
     assert abs(scale.item() - true_amax * 0.999 / 448.0) < 1e-6
+
+
+def test_quant_fp8_scale_used_snapshot():
+    """The epilogue must snapshot the scale it quantized WITH into
+    scale_used before overwriting scale with the next call's value."""
+    from comfyui_parallelanything_amd import ops
+
+    if not ops.hip_available("quant_fp8"):
+        pytest.skip("no quant_fp8 in extension")
+    torch.manual_seed(3)
+    x = torch.randn(128, 1024, device="cuda", dtype=torch.bfloat16)
+    s0 = 0.003  # deliberately NOT amax/448, so next-scale differs from used
+    scale = torch.tensor([s0], device="cuda")
+    amax = torch.zeros(2, device="cuda")
+    used = torch.zeros(1, device="cuda")
+    x8 = ops.quant_fp8(x, scale, amax, scale_used=used)
+    assert abs(used.item() - s0) < 1e-9, "scale_used must be the entry scale"
+    true_amax = x.abs().amax().item()
+    assert abs(scale.item() - true_amax * 0.999 / 448.0) < 1e-6
+    # dequantizing with the USED scale reproduces x (modulo fp8 rounding)
+    deq = x8.float() * used
+    rel = (deq - x.float()).norm() / x.float().norm()
+    assert rel < 0.05
+
+
+def test_fp8_linear_amax_jump_uses_snapshot_scale():
+    """Activation amplitude jumping between calls must not corrupt the
+    output: _scaled_mm dequantizes with the scale actually used, not the
+    next-call scale the kernel epilogue writes (the aliased-buffer bug
+    multiplied the output by next_scale/used_scale)."""
+    from comfyui_parallelanything_amd.models.quant import (
+        FP8Linear, _supports_scaled_mm,
+    )
+
+    if not _supports_scaled_mm():
+        pytest.skip("no fp8 _scaled_mm")
+    torch.manual_seed(4)
+    lin = torch.nn.Linear(1024, 1024).cuda().to(torch.bfloat16)
+    q = FP8Linear.from_linear(lin)
+    x1 = torch.randn(64, 1024, device="cuda", dtype=torch.bfloat16)
+    q(x1)  # warm: sets the delayed scale from x1's amax
+    # 1.5x amplitude: mild clamping only (inherent delayed-scaling cost is
+    # small) but with the bug the output comes back ~1.5x too large.
+    x2 = x1 * 1.5
+    out = q(x2).float()
+    ref = lin(x2).float()
+    rel = (out - ref).norm() / ref.norm()
+    assert rel < 0.10, f"fp8 amax-jump error {rel:.4f}"

@@ -168,9 +168,11 @@ def test_attn_fwd_cross_lengths():
     _cmp(out, ref, 2e-2, 2e-2, "attn cross")
 
 
-def test_qk_norm_rope_fused():
+@pytest.mark.parametrize("D", [64, 128])
+def test_qk_norm_rope_fused(D):
+    """D=64 exercises the lanes>=pairs inactive-lane path (SD3-class heads)."""
     torch.manual_seed(5)
-    B, S, H, D = 2, 100, 4, 128
+    B, S, H = 2, 100, 4
     qkv = torch.randn(B, S, 3, H, D, device="cuda", dtype=torch.bfloat16)
     q, k, v = qkv.unbind(2)
     q_ref = q.clone().permute(0, 2, 1, 3).float()
@@ -181,8 +183,8 @@ def test_qk_norm_rope_fused():
     ops.qk_norm_rope_(q, k, wq, wk, cs)
     ref_q = R.rope_apply(R.rms_norm(q_ref, wq.float()), cs)
     ref_k = R.rope_apply(R.rms_norm(k_ref, wk.float()), cs)
-    _cmp(q.permute(0, 2, 1, 3), ref_q, 3e-2, 3e-2, "fused qk q")
-    _cmp(k.permute(0, 2, 1, 3), ref_k, 3e-2, 3e-2, "fused qk k")
+    _cmp(q.permute(0, 2, 1, 3), ref_q, 3e-2, 3e-2, f"fused qk q D={D}")
+    _cmp(k.permute(0, 2, 1, 3), ref_k, 3e-2, 3e-2, f"fused qk k D={D}")
 
 
 def test_model_forward_gpu_tiny():
@@ -222,9 +224,10 @@ def test_attn_pad_head_dims():
         _cmp(out, ref, 2e-2, 2e-2, f"attn pad D={D}")
 
 
-def test_pack_joint_qkv_fused():
+@pytest.mark.parametrize("D", [64, 128])
+def test_pack_joint_qkv_fused(D):
     torch.manual_seed(7)
-    B, T, Si, H, D = 2, 16, 48, 4, 128
+    B, T, Si, H = 2, 16, 48, 4
     txt_qkv = torch.randn(B, T, 3, H, D, device="cuda", dtype=torch.bfloat16)
     img_qkv = torch.randn(B, Si, 3, H, D, device="cuda", dtype=torch.bfloat16)
     wq_t = torch.randn(D, device="cuda", dtype=torch.bfloat16)
