@@ -102,7 +102,7 @@ def main(argv=None):
         sizes = compute_split_sizes(args.batch, [w / total_w for w in ws])
     else:
         sizes = compute_split_sizes(args.batch, [1.0 / n] * n)
-    if args.model == "wan":
+    if args.model.startswith("wan"):
         x, t, ctx, kw = make_inputs(args.batch, dev=dev, dtype=dtype, tiny=tiny)
     else:
         x, t, ctx, kw = make_inputs(args.batch, px=args.px, dev=dev, dtype=dtype,
@@ -217,6 +217,7 @@ def main(argv=None):
                     "sd15": "SD1.5-class UNet",
                     "sd3": "SD3.5-Large-class MMDiT 8B",
                     "wan": "WAN2.2-class video DiT 14B",
+                    "wan_i2v": "WAN2.2-class I2V video DiT 14B",
                 }[args.model] + (" [TINY DEBUG CONFIG]" if tiny else ""),
                 "global_batch": args.batch,
                 "resolution": args.px,

@@ -144,6 +144,37 @@ def wan_inputs(batch: int, frames: int = 21, h: int = 90, w: int = 160,
     return x, t, ctx, {}
 
 
+def make_wan_i2v(dev="cpu", dtype=torch.bfloat16, tiny=False):
+    cfg = WanConfig.tiny_i2v() if tiny else WanConfig.wan22_a14b_i2v()
+    torch.manual_seed(0)
+    with torch.device(dev):
+        m = WanDiT(cfg)
+    return m.to(dtype=dtype).eval()
+
+
+def wan_i2v_inputs(batch: int, frames: int = 21, h: int = 90, w: int = 160,
+                   dev="cpu", dtype=torch.bfloat16, tiny=False,
+                   seed: int = 1234):
+    """WAN2.2 I2V (BASELINE config 5: 720p batch 4): text-to-video inputs
+    plus the channel-concatenated image conditioning — first-frame mask +
+    reference-image VAE latent, zero on later frames."""
+    cfg = WanConfig.tiny_i2v() if tiny else WanConfig.wan22_a14b_i2v()
+    g = torch.Generator(device="cpu").manual_seed(seed)
+    if tiny:
+        frames, h, w = 4, 8, 8
+    x = torch.randn(batch, cfg.in_channels, frames, h, w, generator=g).to(dev, dtype)
+    t = torch.rand(batch, generator=g).to(dev, torch.float32)
+    txt_len = 512 if not tiny else 8
+    ctx = torch.randn(batch, txt_len, cfg.ctx_dim, generator=g).to(dev, dtype)
+    mask_ch = cfg.cond_channels - cfg.in_channels
+    cond = torch.zeros(batch, cfg.cond_channels, frames, h, w)
+    cond[:, :mask_ch, 0] = 1.0  # first-frame mask
+    cond[:, mask_ch:, 0] = torch.randn(
+        batch, cfg.in_channels, h, w, generator=g
+    )  # reference-image latent on frame 0
+    return x, t, ctx, {"image_cond": cond.to(dev, dtype)}
+
+
 MODELS: Dict[str, Tuple[Callable, Callable]] = {
     "flux": (make_flux, flux_inputs),
     "zimage": (make_zimage, zimage_inputs),
@@ -151,4 +182,5 @@ MODELS: Dict[str, Tuple[Callable, Callable]] = {
     "sdxl": (make_sdxl, sdxl_inputs),
     "sd3": (make_sd3, sd3_inputs),
     "wan": (make_wan, wan_inputs),
+    "wan_i2v": (make_wan_i2v, wan_i2v_inputs),
 }

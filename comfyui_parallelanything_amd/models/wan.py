@@ -80,6 +80,11 @@ class WanBlock(nn.Module):
 @dataclass
 class WanConfig:
     in_channels: int = 16
+    # I2V conditioning channels concatenated on the channel axis BEFORE
+    # patchify: a binary first-frame mask (4ch) + the reference image's VAE
+    # latent (16ch), zero elsewhere in time — the WAN2.2 I2V input layout
+    # (in_dim 36 = 16 noise + 20 cond). 0 = T2V.
+    cond_channels: int = 0
     patch_size: Tuple[int, int, int] = (1, 2, 2)  # (frame, h, w)
     dim: int = 5120
     ffn_dim: int = 13824
@@ -95,6 +100,10 @@ class WanConfig:
         return cls()
 
     @classmethod
+    def wan22_a14b_i2v(cls) -> "WanConfig":
+        return cls(cond_channels=20)
+
+    @classmethod
     def wan22_5b(cls) -> "WanConfig":
         return cls(dim=3072, ffn_dim=14336, num_heads=24, depth=30,
                    patch_size=(1, 2, 2), axes_dim=(44, 42, 42))
@@ -103,6 +112,12 @@ class WanConfig:
     def tiny(cls) -> "WanConfig":
         return cls(in_channels=4, dim=64, ffn_dim=128, num_heads=4, depth=2,
                    ctx_dim=32, axes_dim=(8, 4, 4), time_embed_dim=32)
+
+    @classmethod
+    def tiny_i2v(cls) -> "WanConfig":
+        return cls(in_channels=4, cond_channels=5, dim=64, ffn_dim=128,
+                   num_heads=4, depth=2, ctx_dim=32, axes_dim=(8, 4, 4),
+                   time_embed_dim=32)
 
 
 class WanDiT(nn.Module):
@@ -113,8 +128,11 @@ class WanDiT(nn.Module):
         cfg = cfg or WanConfig()
         self.cfg = cfg
         pf, ph, pw = cfg.patch_size
+        # output predicts noise for the latent channels only; the input
+        # embedding additionally sees the I2V conditioning channels
         self.patch_dim = cfg.in_channels * pf * ph * pw
-        self.patch_in = nn.Linear(self.patch_dim, cfg.dim)
+        self.patch_dim_in = (cfg.in_channels + cfg.cond_channels) * pf * ph * pw
+        self.patch_in = nn.Linear(self.patch_dim_in, cfg.dim)
         self.txt_in = FusedMLP(cfg.ctx_dim, cfg.dim, cfg.dim)
         self.time_in = MLPEmbedder(cfg.time_embed_dim, cfg.dim)
         self.time_proj = nn.Linear(cfg.dim, cfg.dim * 6)
@@ -135,15 +153,33 @@ class WanDiT(nn.Module):
         return pe
 
     @torch.no_grad()
-    def forward(self, x, timesteps, context=None, **kwargs):
+    def forward(self, x, timesteps, context=None, image_cond=None, **kwargs):
         cfg = self.cfg
         B, C, F, H, W = x.shape
         pf, ph, pw = cfg.patch_size
         f, h, w = F // pf, H // ph, W // pw
+        if cfg.cond_channels:
+            # I2V: concatenate mask + reference-image latent channels
+            # (batch-shaped kwarg, so the engine's kwargs-split rules
+            # scatter it with the latent — split.py split_kwargs)
+            if image_cond is None:
+                raise ValueError(
+                    "I2V WanDiT requires image_cond "
+                    f"[B,{cfg.cond_channels},F,H,W]"
+                )
+            if image_cond.shape != (B, cfg.cond_channels, F, H, W):
+                raise ValueError(
+                    f"image_cond shape {tuple(image_cond.shape)} != "
+                    f"{(B, cfg.cond_channels, F, H, W)}"
+                )
+            x_in = torch.cat([x, image_cond.to(x.dtype)], dim=1)
+        else:
+            x_in = x
+        Ct = x_in.shape[1]
         tokens = (
-            x.view(B, C, f, pf, h, ph, w, pw)
+            x_in.view(B, Ct, f, pf, h, ph, w, pw)
             .permute(0, 2, 4, 6, 1, 3, 5, 7)
-            .reshape(B, f * h * w, self.patch_dim)
+            .reshape(B, f * h * w, self.patch_dim_in)
         )
         seq = self.patch_in(tokens)
         if context is None:
