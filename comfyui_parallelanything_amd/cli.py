@@ -46,6 +46,8 @@ def main(argv=None):
     ap.add_argument("--lora", default=None, metavar="PATH[:SCALE]",
                     help="merge a LoRA .safetensors into the weights before "
                          "replication (PEFT or kohya key conventions)")
+    ap.add_argument("--json-out", default=None,
+                    help="write the run summary as one JSON record")
     args = ap.parse_args(argv)
 
     devices = args.devices.split(",")
@@ -82,7 +84,7 @@ def main(argv=None):
     configure_pipeline(engine, microbatches=args.microbatches)
     install_parallel_forward(model, engine)
 
-    if args.model == "wan":
+    if args.model.startswith("wan"):
         x, t, ctx, kw = make_inputs(args.batch, dev=chain.lead, dtype=dtype,
                                     tiny=tiny)
     else:
@@ -102,7 +104,18 @@ def main(argv=None):
             s = timer.stop(args.batch)
             print(f"step {i:3d}: {s.wall_s*1000:8.1f} ms  "
                   f"{s.images_per_s:8.2f} img/s")
-    print(json.dumps(timer.summary(), indent=2))
+    summary = timer.summary()
+    summary["config"] = {
+        "model": args.model, "batch": args.batch, "px": args.px,
+        "devices": devices, "percent": pcts,
+        "dtype": str(dtype).replace("torch.", ""), "tiny": tiny,
+        "engine": "in-process",
+    }
+    print(json.dumps(summary, indent=2))
+    if args.json_out:
+        with open(args.json_out, "w") as f:
+            json.dump(summary, f)
+            f.write("\n")
     cleanup_parallel_model(model)
 
 
