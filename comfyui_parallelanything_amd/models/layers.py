@@ -80,6 +80,26 @@ class MLPEmbedder(nn.Module):
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         return self.out_layer(torch.nn.functional.silu(self.in_layer(x)))
 
+    def forward_timestep(self, t: torch.Tensor,
+                         max_period: float = 10000.0,
+                         time_factor: float = 1000.0) -> torch.Tensor:
+        """Timestep scalar [B] -> conditioning vector [B, hidden].
+
+        GPU: fused sinusoid+in_layer+SiLU kernel (SURVEY §2b timestep-MLP
+        fusion; one launch instead of three and the sinusoid stays fp32
+        into the accumulate), then the out_layer GEMM. CPU / non-bf16:
+        composed ops."""
+        w = self.in_layer.weight
+        if (t.is_cuda and w.dtype == torch.bfloat16
+                and w.shape[1] % 8 == 0
+                and ops.hip_available("timestep_embed_mlp")):
+            h = ops.timestep_embed_mlp(t, w, self.in_layer.bias,
+                                       max_period, time_factor)
+            return self.out_layer(h)
+        emb = ops.timestep_embedding(t, self.in_layer.in_features,
+                                     max_period, time_factor)
+        return self.forward(emb.to(w.dtype))
+
 
 class RMSNorm(nn.Module):
     def __init__(self, dim: int):

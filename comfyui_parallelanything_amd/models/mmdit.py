@@ -149,18 +149,14 @@ class Flux(nn.Module):
         if context is None:
             context = torch.zeros(B, 1, cfg.context_dim, device=x.device, dtype=x.dtype)
         txt = self.txt_in(context)
-        vec = self.time_in(
-            ops.timestep_embedding(timesteps, cfg.time_embed_dim).to(x.dtype)
-        )
+        vec = self.time_in.forward_timestep(timesteps)
         if y is None:
             y = torch.zeros(B, cfg.vec_dim, device=x.device, dtype=x.dtype)
         vec = vec + self.vector_in(y)
         if self.guidance_in is not None:
             if guidance is None:
                 guidance = torch.full_like(timesteps, 4.0)
-            vec = vec + self.guidance_in(
-                ops.timestep_embedding(guidance, cfg.time_embed_dim).to(x.dtype)
-            )
+            vec = vec + self.guidance_in.forward_timestep(guidance)
         pe = self._pe(h, w, txt.shape[1], x.device, x.dtype)
 
         bank = self._mod_cache.get("bank")
@@ -263,9 +259,7 @@ class ZImage(nn.Module):
         if context is None:
             context = torch.zeros(B, 1, cfg.context_dim, device=x.device, dtype=x.dtype)
         txt = self.txt_in(context)
-        vec = self.time_in(
-            ops.timestep_embedding(timesteps, cfg.time_embed_dim).to(x.dtype)
-        )
+        vec = self.time_in.forward_timestep(timesteps)
         seq = torch.cat([txt, img], dim=1)
         pe = self._pe(h, w, txt.shape[1], x.device)
         bank = self._mod_cache.get("bank")

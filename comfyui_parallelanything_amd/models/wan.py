@@ -185,9 +185,7 @@ class WanDiT(nn.Module):
         if context is None:
             context = torch.zeros(B, 1, cfg.ctx_dim, device=x.device, dtype=x.dtype)
         ctx = self.txt_in(context)
-        tvec = self.time_in(
-            ops.timestep_embedding(timesteps, cfg.time_embed_dim).to(x.dtype)
-        )
+        tvec = self.time_in.forward_timestep(timesteps)
         e = self.time_proj(torch.nn.functional.silu(tvec)).view(B, 6, cfg.dim)
         pe = self._pe(f, h, w, x.device)
         for block in self.transformer_blocks:

@@ -298,3 +298,15 @@ def timestep_embedding(t, dim: int, max_period: float = 10000.0,
             return ext.timestep_embedding(t.contiguous().float(), int(dim),
                                           float(max_period), float(time_factor))
     return reference.timestep_embedding(t, dim, max_period, time_factor)
+
+
+def timestep_embed_mlp(t, w1, b1, max_period: float = 10000.0,
+                       time_factor: float = 1000.0) -> torch.Tensor:
+    """Fused sinusoidal embedding + first Linear + SiLU: [B] -> [B, H].
+
+    SURVEY §2b timestep-MLP fusion (the in_layer half; the out Linear is a
+    GEMM and stays on hipBLASLt). GPU-only — callers fall back to the
+    composed path on CPU (MLPEmbedder.forward_timestep)."""
+    ext = _require_ext("timestep_embed_mlp")
+    return ext.timestep_embed_mlp(t, w1, b1, float(max_period),
+                                  float(time_factor))

@@ -400,6 +400,51 @@ __global__ void timestep_embedding_kernel(const float* __restrict__ t,
 
 
 // ---------------------------------------------------------------------------
+// Fused timestep-embedding MLP input half: out = silu(sinusoid(t) @ W1^T + b1)
+// (SURVEY §2b "sinusoidal + 2xLinear+SiLU fused"; the out_layer Linear is a
+// GEMM-shaped op and stays on hipBLASLt). One 256-thread block covers 256
+// hidden outputs of one batch element; the K-dim sinusoid is computed once
+// into LDS per block. B is tiny (<=32) so this stage is launch-bound —
+// collapsing embed+linear+silu into one kernel removes two launches and the
+// [B,K] fp32 + bf16-cast intermediates (and keeps the sinusoid fp32 into
+// the accumulate, slightly MORE accurate than the cast-then-GEMM path).
+// ---------------------------------------------------------------------------
+__global__ void ts_embed_mlp_kernel(const float* __restrict__ t,
+                                    const bf16* __restrict__ w1,  // [H, K]
+                                    const bf16* __restrict__ b1,  // [H]|null
+                                    bf16* __restrict__ out,       // [B, H]
+                                    int H, int K,
+                                    float max_period, float time_factor) {
+    extern __shared__ float emb[];  // [K]
+    const int hblocks = (H + 255) / 256;
+    const int b = blockIdx.x / hblocks;
+    const int hblk = blockIdx.x % hblocks;
+    const int half = K / 2;
+    const float tv = t[b] * time_factor;
+    for (int j = (int)threadIdx.x; j < half; j += (int)blockDim.x) {
+        const float freq = __expf(-__logf(max_period) * (float)j / (float)half);
+        const float arg = tv * freq;
+        emb[j] = __cosf(arg);
+        emb[half + j] = __sinf(arg);
+    }
+    __syncthreads();
+    const int o = hblk * 256 + (int)threadIdx.x;
+    if (o >= H) return;
+    float acc = b1 ? bf2f(b1[o]) : 0.f;
+    const short8* wrow = reinterpret_cast<const short8*>(w1 + (long)o * K);
+    for (int k8 = 0; k8 < K / 8; ++k8) {
+        const short8 w = wrow[k8];
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+            acc += emb[k8 * 8 + j] *
+                   bf2f(__ushort_as_bfloat16((unsigned short)w[j]));
+    }
+    acc = acc / (1.f + __expf(-acc));  // SiLU
+    out[(long)b * H + o] = f2bf(acc);
+}
+
+
+// ---------------------------------------------------------------------------
 // Fused qk-norm + RoPE (in-place): per (b, s, h) row of q and k,
 // RMSNorm over D with per-head-dim weight, then rotary by cs[s].
 // One wave handles one (b, s, h) row of BOTH q and k — single pass over the
@@ -1026,6 +1071,37 @@ at::Tensor timestep_embedding(at::Tensor t, long dim, double max_period,
     hipLaunchKernelGGL(timestep_embedding_kernel, dim3(std::max(blocks, 1)),
                        dim3(256), 0, cur_stream(), tc.data_ptr<float>(),
                        out.data_ptr<float>(), B, (int)dim, (float)max_period,
+                       (float)time_factor);
+    return out;
+}
+
+
+at::Tensor timestep_embed_mlp(at::Tensor t, at::Tensor w1,
+                              c10::optional<at::Tensor> b1,
+                              double max_period, double time_factor) {
+    CHECK_GPU(t);
+    TORCH_CHECK(w1.scalar_type() == at::kBFloat16,
+                "timestep_embed_mlp: bf16 weight");
+    auto tc = t.to(at::kFloat).contiguous();
+    auto w1c = w1.contiguous();
+    const int H = (int)w1c.size(0), K = (int)w1c.size(1);
+    TORCH_CHECK(K % 8 == 0 && K <= 4096,
+                "timestep_embed_mlp: K % 8 == 0 and K <= 4096");
+    const int B = (int)tc.numel();
+    auto out = at::empty({B, H}, w1c.options());
+    const bf16* bias = nullptr;
+    at::Tensor b1c;
+    if (b1.has_value()) {
+        b1c = b1->contiguous();
+        TORCH_CHECK(b1c.scalar_type() == at::kBFloat16 &&
+                    (int)b1c.numel() == H, "timestep_embed_mlp: bias [H] bf16");
+        bias = (const bf16*)b1c.data_ptr();
+    }
+    const int hblocks = (H + 255) / 256;
+    hipLaunchKernelGGL(ts_embed_mlp_kernel, dim3(B * hblocks), dim3(256),
+                       K * sizeof(float), cur_stream(), tc.data_ptr<float>(),
+                       (const bf16*)w1c.data_ptr(), bias,
+                       (bf16*)out.data_ptr(), H, K, (float)max_period,
                        (float)time_factor);
     return out;
 }
@@ -1684,6 +1760,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("gelu_tanh", &gelu_tanh, "Vectorized tanh-GELU (gfx950)");
     m.def("quant_fp8", &quant_fp8,
           "Fused bf16->e4m3fn quant with running amax (gfx950)");
+    m.def("timestep_embed_mlp", &timestep_embed_mlp,
+          "Fused sinusoidal embed + Linear + SiLU (gfx950)");
     m.def("pack_joint_qkv", &pack_joint_qkv,
           "Fused dual-stream qkv pack + qk-norm + RoPE (gfx950)");
     m.def("qk_norm_rope_", &qk_norm_rope_,

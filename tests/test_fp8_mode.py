@@ -115,10 +115,11 @@ def test_quant_fp8_scale_used_snapshot():
     x = torch.randn(128, 1024, device="cuda", dtype=torch.bfloat16)
     s0 = 0.003  # deliberately NOT amax/448, so next-scale differs from used
     scale = torch.tensor([s0], device="cuda")
+    s0_f32 = scale.item()  # the fp32 value the kernel actually reads
     amax = torch.zeros(2, device="cuda")
     used = torch.zeros(1, device="cuda")
     x8 = ops.quant_fp8(x, scale, amax, scale_used=used)
-    assert abs(used.item() - s0) < 1e-9, "scale_used must be the entry scale"
+    assert used.item() == s0_f32, "scale_used must be the entry scale bits"
     true_amax = x.abs().amax().item()
     assert abs(scale.item() - true_amax * 0.999 / 448.0) < 1e-6
     # dequantizing with the USED scale reproduces x (modulo fp8 rounding)
