@@ -182,6 +182,15 @@ def attention_bshd(q, k, v, scale: Optional[float] = None) -> torch.Tensor:
     Head dims that are not 64/128 but fit under them (SD1.5's 40/80) are
     zero-padded: padding q/k leaves every softmax score unchanged, padded v
     columns are sliced off the output.
+
+    Cost statement: the pad wastes attention FLOPs
+    proportional to Dp/D — 60% at D=40->64, 60% at D=80->128 — plus one
+    pad copy of q/k/v per call. This is acceptable because SD1.5 is the
+    PLUMBING config (BASELINE config 1 runs it on cpu,cpu); no SD-class
+    GPU performance is claimed anywhere. A native D=48/96 tile would need
+    fractional per-thread stage vectors (KVBLK*D/8 not divisible by 512
+    threads) — not worth the schedule split while no headline depends on
+    these shapes.
     """
     if scale is None:
         scale = 1.0 / math.sqrt(q.shape[-1])

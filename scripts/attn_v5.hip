@@ -1,0 +1,460 @@
+// v5 deep-pipelined attention schedule A/B harness (gfx950).
+//
+// Round-1 verdict: v4 sits AT the 2-barrier structural ceiling (~900 TF
+// flux). The way past (guide §5.5 regime gate) is a deeper pipeline:
+// double-buffered LDS so the K/V stage for tile t+1 overlaps compute of
+// tile t with ONE barrier per tile instead of two, stores interleaved
+// into the compute stream where the target pipe is idle.
+//
+// Variants (within-probe interleaved, guide rule 24):
+//   0: shipped v4 (single LDS buffer, 2 barriers/tile)
+//   1: v5a — double-buffered LDS, 1 barrier/tile; store order:
+//      QK^T -> K-store -> softmax -> V-store -> next-loads -> Pcvt -> PV
+//   2: v5b — double-buffered LDS, 1 barrier/tile; stores first:
+//      K+V-store -> QK^T -> softmax -> next-loads -> Pcvt -> PV
+//
+// Correctness gates (guide rules 24/26): full-tensor vs var0 on a tail
+// shape at D=128 and D=64, a K-row spike forcing late rescale, and a
+// bitwise double-run race screen per variant.
+//
+// Build: hipcc --offload-arch=gfx950 -O3 -std=c++17 attn_v5.hip -o attn_v5
+// Run:   ./attn_v5 [rounds]
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <cstdio>
+#include <cstdlib>
+#include <vector>
+#include <algorithm>
+#include <cmath>
+
+using bf16 = __hip_bfloat16;
+typedef __attribute__((ext_vector_type(8))) short bf16x8;
+typedef __attribute__((ext_vector_type(4))) short s16x4;
+typedef __attribute__((ext_vector_type(16))) float f32x16;
+#define LDS_P __attribute__((address_space(3)))
+#define PA_DEV __device__ __forceinline__
+#define PA_LOG2E 1.4426950408889634f
+
+PA_DEV float bf2f(bf16 v) { return __bfloat162float(v); }
+PA_DEV bf16 f2bf(float v) { return __float2bfloat16(v); }
+PA_DEV f32x16 mfma32x32x16(bf16x8 a, bf16x8 b, f32x16 c) {
+    return __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, c, 0, 0, 0);
+}
+PA_DEV unsigned int cvt_pk_bf16(float lo, float hi) {
+    unsigned int r;
+    asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(r) : "v"(lo), "v"(hi));
+    return r;
+}
+
+// VAR: 0 = v4 (DBUF 1, 2 barriers); 1 = v5a; 2 = v5b
+template <int D, int VAR>
+__global__ __launch_bounds__(512, 2) void attn_v5_kernel(
+    const bf16* __restrict__ q, const bf16* __restrict__ k,
+    const bf16* __restrict__ v, bf16* __restrict__ out,
+    int S, int Sk, float scale, int H) {
+    constexpr int KVBLK = 64;
+    constexpr int WAVES = 8;
+    constexpr int THREADS = WAVES * 64;
+    constexpr int KPAD = D + 8;
+    constexpr int VROW = 160;          // 40 granules ≡ 8 mod 32
+    constexpr int KK = D / 16;
+    constexpr int NV = D / 32;
+    constexpr int KVECS = (KVBLK * D) / (8 * THREADS);
+    constexpr int DBUF = (VAR == 0) ? 1 : 2;
+
+    __shared__ bf16 k_lds[DBUF * KVBLK * KPAD];
+    __shared__ bf16 v_lds[DBUF * KVBLK * VROW];
+
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wid = tid >> 6;
+    const int l32 = lane & 31;
+    const int hi = lane >> 5;
+    const int trb = (hi * 8 + ((lane & 15) >> 2)) * VROW +
+                    16 * (((lane >> 4) & 1) ^ hi) + 4 * (lane & 3);
+
+    const int nq = (S + WAVES * 32 - 1) / (WAVES * 32);
+    const long id = blockIdx.x;
+    const long bh = (id & 7) + 8 * ((id >> 3) / nq);
+    const int qtile = (int)((id >> 3) % nq);
+    const long b = bh / H;
+    const int h = (int)(bh % H);
+    const int q0 = qtile * (WAVES * 32) + wid * 32;
+    const int ss = H * D;
+
+    const bf16* qp = q + (b * (long)S + 0) * ss + (long)h * D;
+    const bf16* kp = k + (b * (long)Sk + 0) * ss + (long)h * D;
+    const bf16* vp = v + (b * (long)Sk + 0) * ss + (long)h * D;
+    bf16* op = out + (b * (long)S + 0) * ss + (long)h * D;
+
+    bf16x8 qfrag[KK];
+    {
+        const int row = q0 + l32;
+        const int rr = row < S ? row : S - 1;
+#pragma unroll
+        for (int kk = 0; kk < KK; ++kk)
+            qfrag[kk] = *reinterpret_cast<const bf16x8*>(
+                qp + (long)rr * ss + kk * 16 + hi * 8);
+    }
+
+    f32x16 o_acc[NV];
+#pragma unroll
+    for (int n = 0; n < NV; ++n)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) o_acc[n][r] = 0.f;
+    float m_run = -1e30f, l_run = 0.f;
+    const float scale2 = scale * PA_LOG2E;
+
+    bf16x8 kreg[KVECS], vreg[KVECS];
+    auto issue_tile_loads = [&](int kv0) {
+#pragma unroll
+        for (int i = 0; i < KVECS; ++i) {
+            const int idx = tid + i * THREADS;
+            const int row = idx / (D / 8);
+            const int col = (idx % (D / 8)) * 8;
+            const int src = kv0 + row;
+            if (src < Sk) {
+                kreg[i] = *reinterpret_cast<const bf16x8*>(
+                    kp + (long)src * ss + col);
+                vreg[i] = *reinterpret_cast<const bf16x8*>(
+                    vp + (long)src * ss + col);
+            } else {
+                kreg[i] = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+                vreg[i] = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+            }
+        }
+    };
+    auto write_k_lds = [&](int buf) {
+#pragma unroll
+        for (int i = 0; i < KVECS; ++i) {
+            const int idx = tid + i * THREADS;
+            const int row = idx / (D / 8);
+            const int col = (idx % (D / 8)) * 8;
+            *reinterpret_cast<bf16x8*>(
+                &k_lds[buf * (KVBLK * KPAD) + row * KPAD + col]) = kreg[i];
+        }
+    };
+    auto write_v_lds = [&](int buf) {
+#pragma unroll
+        for (int i = 0; i < KVECS; ++i) {
+            const int idx = tid + i * THREADS;
+            const int row = idx / (D / 8);
+            const int col = (idx % (D / 8)) * 8;
+            *reinterpret_cast<bf16x8*>(
+                &v_lds[buf * (KVBLK * VROW) + row * VROW +
+                       (col ^ ((row & 8) << 1))]) = vreg[i];
+        }
+    };
+
+    auto qk_half = [&](int buf, f32x16* st) {
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int kt = 0; kt < 2; ++kt) {
+#pragma unroll
+            for (int r = 0; r < 16; ++r) st[kt][r] = 0.f;
+#pragma unroll
+            for (int kk = 0; kk < KK; ++kk) {
+                bf16x8 afrag = *reinterpret_cast<const bf16x8*>(
+                    &k_lds[buf * (KVBLK * KPAD) + (kt * 32 + l32) * KPAD +
+                           kk * 16 + hi * 8]);
+                st[kt] = mfma32x32x16(afrag, qfrag[kk], st[kt]);
+            }
+        }
+        __builtin_amdgcn_s_setprio(0);
+    };
+    auto softmax_half = [&](int kv0, f32x16* st, bf16x8* pfrag) {
+        float mx = -3e30f;
+#pragma unroll
+        for (int kt = 0; kt < 2; ++kt)
+#pragma unroll
+            for (int r = 0; r < 16; ++r) {
+                const int key = kv0 + kt * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+                const float sv = (key < Sk) ? st[kt][r] : -3e30f;
+                st[kt][r] = sv;
+                mx = fmaxf(mx, sv);
+            }
+        mx = fmaxf(mx, __shfl_xor(mx, 32, 64));
+        const float mnew = fmaxf(m_run, mx * scale2);
+        const float alpha = __builtin_amdgcn_exp2f(m_run - mnew);
+        m_run = mnew;
+        float ps = 0.f;
+#pragma unroll
+        for (int kt = 0; kt < 2; ++kt)
+#pragma unroll
+            for (int r = 0; r < 16; ++r) {
+                const float pv_ =
+                    __builtin_amdgcn_exp2f(fmaf(st[kt][r], scale2, -mnew));
+                st[kt][r] = pv_;
+                ps += pv_;
+            }
+        ps += __shfl_xor(ps, 32, 64);
+        l_run = l_run * alpha + ps;
+        if (alpha != 1.f) {
+#pragma unroll
+            for (int n = 0; n < NV; ++n)
+#pragma unroll
+                for (int r = 0; r < 16; ++r) o_acc[n][r] *= alpha;
+        }
+#pragma unroll
+        for (int c = 0; c < 4; ++c) {
+            const f32x16& sv = st[c >> 1];
+            const int rb = 8 * (c & 1);
+            unsigned int w0 = cvt_pk_bf16(sv[rb + 0], sv[rb + 1]);
+            unsigned int w1 = cvt_pk_bf16(sv[rb + 2], sv[rb + 3]);
+            unsigned int w2 = cvt_pk_bf16(sv[rb + 4], sv[rb + 5]);
+            unsigned int w3 = cvt_pk_bf16(sv[rb + 6], sv[rb + 7]);
+            auto r02 = __builtin_amdgcn_permlane32_swap(w0, w2, false, false);
+            auto r13 = __builtin_amdgcn_permlane32_swap(w1, w3, false, false);
+            unsigned int d[4] = {(unsigned int)r02[0], (unsigned int)r13[0],
+                                 (unsigned int)r02[1], (unsigned int)r13[1]};
+            pfrag[c] = *reinterpret_cast<bf16x8*>(d);
+        }
+    };
+    auto pv_half = [&](int buf, bf16x8* pfrag) {
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int c = 0; c < 4; ++c) {
+#pragma unroll
+            for (int n = 0; n < NV; ++n) {
+                s16x4 alo = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+                    (LDS_P s16x4*)&v_lds[buf * (KVBLK * VROW) + trb +
+                                         c * (16 * VROW) + n * 32]);
+                s16x4 ahi = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+                    (LDS_P s16x4*)&v_lds[buf * (KVBLK * VROW) + trb +
+                                         c * (16 * VROW) + 4 * VROW + n * 32]);
+                bf16x8 va =
+                    __builtin_shufflevector(alo, ahi, 0, 1, 2, 3, 4, 5, 6, 7);
+                o_acc[n] = mfma32x32x16(va, pfrag[c], o_acc[n]);
+            }
+        }
+        __builtin_amdgcn_s_setprio(0);
+    };
+
+    const int n_tiles = (Sk + KVBLK - 1) / KVBLK;
+    if (VAR == 0) {
+        issue_tile_loads(0);
+        for (int t = 0; t < n_tiles; ++t) {
+            __syncthreads();
+            write_k_lds(0);
+            write_v_lds(0);
+            __syncthreads();
+            if (t + 1 < n_tiles) issue_tile_loads((t + 1) * KVBLK);
+            f32x16 st[2];
+            bf16x8 pfrag[4];
+            qk_half(0, st);
+            softmax_half(t * KVBLK, st, pfrag);
+            pv_half(0, pfrag);
+        }
+    } else {
+        // v5: double-buffered LDS, one barrier per tile. Prologue stages
+        // tile 0 into buf 0 and leaves tile 1 in registers.
+        issue_tile_loads(0);
+        write_k_lds(0);
+        write_v_lds(0);
+        if (n_tiles > 1) issue_tile_loads(KVBLK);
+        __syncthreads();
+        for (int t = 0; t < n_tiles; ++t) {
+            const int p = t & 1;
+            f32x16 st[2];
+            bf16x8 pfrag[4];
+            if (VAR == 2) {
+                // v5b: stores first (the whole stage ahead of compute)
+                if (t + 1 < n_tiles) {
+                    write_k_lds(p ^ 1);
+                    write_v_lds(p ^ 1);
+                }
+                qk_half(p, st);
+                softmax_half(t * KVBLK, st, pfrag);
+                if (t + 2 < n_tiles) issue_tile_loads((t + 2) * KVBLK);
+            } else {
+                // v5a: stores interleaved where the LDS port is idle
+                qk_half(p, st);
+                if (t + 1 < n_tiles) write_k_lds(p ^ 1);
+                softmax_half(t * KVBLK, st, pfrag);
+                if (t + 1 < n_tiles) write_v_lds(p ^ 1);
+                if (t + 2 < n_tiles) issue_tile_loads((t + 2) * KVBLK);
+            }
+            pv_half(p, pfrag);
+            __syncthreads();
+        }
+    }
+
+    const int row = q0 + l32;
+    if (row < S) {
+        const float inv_l = (l_run > 0.f) ? 1.f / l_run : 0.f;
+#pragma unroll
+        for (int n = 0; n < NV; ++n)
+#pragma unroll
+            for (int r2 = 0; r2 < 4; ++r2) {
+                const int dim0 = n * 32 + 8 * r2 + 4 * hi;
+                unsigned short pack[4];
+#pragma unroll
+                for (int j = 0; j < 4; ++j)
+                    pack[j] = __bfloat16_as_ushort(
+                        f2bf(o_acc[n][r2 * 4 + j] * inv_l));
+                *reinterpret_cast<unsigned long long*>(
+                    op + (long)row * ss + dim0) =
+                    *reinterpret_cast<unsigned long long*>(pack);
+            }
+    }
+}
+
+#define HIP_CHECK(x) do { hipError_t e = (x); if (e) { \
+    printf("HIP error %d at %d\n", e, __LINE__); exit(1); } } while (0)
+
+static void fill_random(bf16* dst, long n, unsigned seed) {
+    std::vector<unsigned short> h(1 << 20);
+    unsigned x = seed;
+    for (auto& e : h) {
+        x = x * 1664525u + 1013904223u;
+        float f = ((x >> 8) / 8388608.0f) * 2.f - 1.f;
+        unsigned int bits;
+        __builtin_memcpy(&bits, &f, 4);
+        e = (unsigned short)(bits >> 16);
+    }
+    for (long off = 0; off < n; off += (1 << 20)) {
+        long len = std::min<long>(1 << 20, n - off);
+        HIP_CHECK(hipMemcpy(dst + off, h.data(), len * 2,
+                            hipMemcpyHostToDevice));
+    }
+}
+
+template <int D>
+static void launch(int var, const bf16* q, const bf16* k, const bf16* v,
+                   bf16* o, int B, int H, int S, float scale) {
+    dim3 grid(((S + 255) / 256) * B * H), blk(512);
+    switch (var) {
+    case 0:
+        hipLaunchKernelGGL((attn_v5_kernel<D, 0>), grid, blk, 0, 0, q, k, v,
+                           o, S, S, scale, H);
+        break;
+    case 1:
+        hipLaunchKernelGGL((attn_v5_kernel<D, 1>), grid, blk, 0, 0, q, k, v,
+                           o, S, S, scale, H);
+        break;
+    default:
+        hipLaunchKernelGGL((attn_v5_kernel<D, 2>), grid, blk, 0, 0, q, k, v,
+                           o, S, S, scale, H);
+    }
+}
+
+constexpr int NVAR = 3;
+
+template <int D>
+static int check_correct(int B, int H, int S, bool spike) {
+    const long n = (long)B * S * H * D;
+    bf16 *q, *k, *v, *o;
+    HIP_CHECK(hipMalloc(&q, n * 2));
+    HIP_CHECK(hipMalloc(&k, n * 2));
+    HIP_CHECK(hipMalloc(&v, n * 2));
+    HIP_CHECK(hipMalloc(&o, n * 2));
+    fill_random(q, n, 1234);
+    fill_random(k, n, 777);
+    fill_random(v, n, 4242);
+    if (spike) {
+        // spike a LATE K row so the running max jumps past tile 0's
+        // (rule 26: force the rescale branch)
+        std::vector<unsigned short> big(H * D, 0x4120);  // 10.0 bf16
+        HIP_CHECK(hipMemcpy(k + (long)(S - 7) * H * D, big.data(),
+                            big.size() * 2, hipMemcpyHostToDevice));
+    }
+    const float scale = 1.0f / sqrtf((float)D);
+    std::vector<unsigned short> ref(n), got(n), got2(n);
+    launch<D>(0, q, k, v, o, B, H, S, scale);
+    HIP_CHECK(hipGetLastError());
+    HIP_CHECK(hipDeviceSynchronize());
+    HIP_CHECK(hipMemcpy(ref.data(), o, n * 2, hipMemcpyDeviceToHost));
+    int fails = 0;
+    for (int var = 1; var < NVAR; ++var) {
+        HIP_CHECK(hipMemset(o, 0, n * 2));
+        launch<D>(var, q, k, v, o, B, H, S, scale);
+        HIP_CHECK(hipGetLastError());
+        HIP_CHECK(hipDeviceSynchronize());
+        HIP_CHECK(hipMemcpy(got.data(), o, n * 2, hipMemcpyDeviceToHost));
+        // race screen: second run must be bitwise identical
+        HIP_CHECK(hipMemset(o, 0, n * 2));
+        launch<D>(var, q, k, v, o, B, H, S, scale);
+        HIP_CHECK(hipDeviceSynchronize());
+        HIP_CHECK(hipMemcpy(got2.data(), o, n * 2, hipMemcpyDeviceToHost));
+        long bad = 0, race = 0;
+        for (long i = 0; i < n; ++i) {
+            if (got[i] != got2[i]) ++race;
+            float a, b2;
+            unsigned int ua = (unsigned int)ref[i] << 16,
+                         ub = (unsigned int)got[i] << 16;
+            __builtin_memcpy(&a, &ua, 4);
+            __builtin_memcpy(&b2, &ub, 4);
+            if (fabsf(a - b2) > 0.05f + 0.05f * fabsf(a)) ++bad;
+        }
+        if (bad || race) {
+            printf("VAR %d WRONG D=%d S=%d spike=%d: %ld/%ld mismatch, "
+                   "%ld race\n", var, D, S, (int)spike, bad, n, race);
+            ++fails;
+        }
+    }
+    (void)hipFree(q); (void)hipFree(k); (void)hipFree(v); (void)hipFree(o);
+    return fails;
+}
+
+int main(int argc, char** argv) {
+    const int rounds = argc > 1 ? atoi(argv[1]) : 8;
+    int fails = 0;
+    fails += check_correct<128>(2, 3, 1000, false);
+    fails += check_correct<128>(2, 3, 1000, true);
+    fails += check_correct<64>(2, 3, 1000, false);
+    fails += check_correct<64>(2, 3, 1000, true);
+    if (fails) { printf("CORRECTNESS FAILED (%d)\n", fails); return 1; }
+    printf("correctness: all variants agree (tails + spike + race screen)\n");
+
+    struct Shape { int B, H, S, D; const char* name; };
+    Shape shapes[3] = {{8, 24, 4608, 128, "flux"},
+                       {1, 8, 30720, 128, "long"},
+                       {8, 38, 4250, 64, "sd3-d64"}};
+    for (auto& sh : shapes) {
+        const int B = sh.B, H = sh.H, S = sh.S, D = sh.D;
+        const long n = (long)B * S * H * D;
+        bf16 *q, *k, *v, *o;
+        HIP_CHECK(hipMalloc(&q, n * 2));
+        HIP_CHECK(hipMalloc(&k, n * 2));
+        HIP_CHECK(hipMalloc(&v, n * 2));
+        HIP_CHECK(hipMalloc(&o, n * 2));
+        fill_random(q, n, 12345);
+        fill_random(k, n, 54321);
+        fill_random(v, n, 999);
+        const float scale = 1.0f / sqrtf((float)D);
+        const double tf = 4.0 * B * H * (double)S * S * D / 1e12;
+        double best[NVAR];
+        for (auto& x : best) x = 1e30;
+        auto run = [&](int var) {
+            if (D == 64)
+                launch<64>(var, q, k, v, o, B, H, S, scale);
+            else
+                launch<128>(var, q, k, v, o, B, H, S, scale);
+        };
+        run(0);
+        HIP_CHECK(hipDeviceSynchronize());
+        for (int r = 0; r < rounds; ++r)
+            for (int var = 0; var < NVAR; ++var) {
+                hipEvent_t e0, e1;
+                HIP_CHECK(hipEventCreate(&e0));
+                HIP_CHECK(hipEventCreate(&e1));
+                HIP_CHECK(hipEventRecord(e0));
+                for (int it = 0; it < 3; ++it) run(var);
+                HIP_CHECK(hipEventRecord(e1));
+                HIP_CHECK(hipEventSynchronize(e1));
+                float ms;
+                HIP_CHECK(hipEventElapsedTime(&ms, e0, e1));
+                if (ms / 3.0 < best[var]) best[var] = ms / 3.0;
+                (void)hipEventDestroy(e0);
+                (void)hipEventDestroy(e1);
+            }
+        printf("%-8s", sh.name);
+        for (int var = 0; var < NVAR; ++var)
+            printf("  v%d %7.3f ms (%6.1f TF)", var, best[var],
+                   tf / best[var] * 1e3);
+        printf("\n");
+        (void)hipFree(q); (void)hipFree(k); (void)hipFree(v); (void)hipFree(o);
+    }
+    return 0;
+}
