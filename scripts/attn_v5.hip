@@ -399,11 +399,14 @@ static int check_correct(int B, int H, int S, bool spike) {
 
 int main(int argc, char** argv) {
     const int rounds = argc > 1 ? atoi(argv[1]) : 8;
+    // NOTE: B*H must be a multiple of 8 — the XCD-affine decode is only
+    // bijective then (the shipped host wrapper gates on this; the harness
+    // just uses conforming shapes). S=1000 exercises the 40-key tail tile.
     int fails = 0;
-    fails += check_correct<128>(2, 3, 1000, false);
-    fails += check_correct<128>(2, 3, 1000, true);
-    fails += check_correct<64>(2, 3, 1000, false);
-    fails += check_correct<64>(2, 3, 1000, true);
+    fails += check_correct<128>(2, 4, 1000, false);
+    fails += check_correct<128>(2, 4, 1000, true);
+    fails += check_correct<64>(2, 4, 1000, false);
+    fails += check_correct<64>(2, 4, 1000, true);
     if (fails) { printf("CORRECTNESS FAILED (%d)\n", fails); return 1; }
     printf("correctness: all variants agree (tails + spike + race screen)\n");
 
