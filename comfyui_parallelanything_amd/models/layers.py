@@ -56,6 +56,17 @@ class GELULinear(nn.Module):
             return out.reshape(*x.shape[:-1], out.shape[-1])
         return torch.nn.functional.gelu(self.lin(x), approximate="tanh")
 
+    def forward_ln(self, x, scale, shift):
+        """AdaLN-modulated LayerNorm feeding this projection; in fp8 mode
+        the quant fuses into the LN kernel (models/quant.py ln_quant)."""
+        ln_fwd = getattr(self.lin, "forward_ln", None)
+        if ln_fwd is not None:
+            y = ln_fwd(x, scale, shift)
+            if y.is_cuda:
+                return ops.gelu_tanh(y)
+            return torch.nn.functional.gelu(y, approximate="tanh")
+        return self.forward(ops.layer_norm_mod(x, scale, shift))
+
 
 class FusedMLP(nn.Module):
     """Linear -> GELU (epilogue-fused) -> Linear."""
@@ -67,6 +78,19 @@ class FusedMLP(nn.Module):
 
     def forward(self, x):
         return self.down(self.up(x))
+
+    def forward_ln(self, x, scale, shift):
+        return self.down(self.up.forward_ln(x, scale, shift))
+
+
+def ln_mod_into(proj, x, scale, shift):
+    """layer_norm_mod -> proj; fp8-aware projections (FP8Linear,
+    GELULinear, FusedMLP) fuse the e4m3 cast INTO the LN kernel via their
+    own forward_ln — one HBM pass instead of LN + standalone quant."""
+    fwd = getattr(proj, "forward_ln", None)
+    if fwd is not None:
+        return fwd(x, scale, shift)
+    return proj(ops.layer_norm_mod(x, scale, shift))
 
 
 class MLPEmbedder(nn.Module):
@@ -224,12 +248,14 @@ class DoubleStreamBlock(nn.Module):
             img_m1, img_m2 = self.img_mod(vec)
             txt_m1, txt_m2 = self.txt_mod(vec)
 
-        img_in = ops.layer_norm_mod(img, img_m1.scale, img_m1.shift)
-        txt_in = ops.layer_norm_mod(txt, txt_m1.scale, txt_m1.shift)
         T = txt.shape[1]
         H = self.num_heads
-        txt_qkv = self.txt_attn_qkv(txt_in).unflatten(-1, (3, H, -1))
-        img_qkv = self.img_attn_qkv(img_in).unflatten(-1, (3, H, -1))
+        txt_qkv = ln_mod_into(
+            self.txt_attn_qkv, txt, txt_m1.scale, txt_m1.shift
+        ).unflatten(-1, (3, H, -1))
+        img_qkv = ln_mod_into(
+            self.img_attn_qkv, img, img_m1.scale, img_m1.shift
+        ).unflatten(-1, (3, H, -1))
         # fused: per-stream qk-norm + RoPE + contiguous joint q/k/v
         # (txt first, then img — FLUX convention)
         q, k, v = ops.pack_joint_qkv(
@@ -249,12 +275,12 @@ class DoubleStreamBlock(nn.Module):
         img = ops.gate_residual(img, img_m1.gate, self.img_attn_proj(img_attn))
         img = ops.gate_residual(
             img, img_m2.gate,
-            self.img_mlp(ops.layer_norm_mod(img, img_m2.scale, img_m2.shift)),
+            self.img_mlp.forward_ln(img, img_m2.scale, img_m2.shift),
         )
         txt = ops.gate_residual(txt, txt_m1.gate, self.txt_attn_proj(txt_attn))
         txt = ops.gate_residual(
             txt, txt_m2.gate,
-            self.txt_mlp(ops.layer_norm_mod(txt, txt_m2.scale, txt_m2.shift)),
+            self.txt_mlp.forward_ln(txt, txt_m2.scale, txt_m2.shift),
         )
         return img, txt
 
@@ -284,9 +310,22 @@ class SingleStreamBlock(nn.Module):
             (mod,) = mods
         else:
             mod, _ = self.modulation(vec)
-        x_in = ops.layer_norm_mod(x, mod.scale, mod.shift)
-        qkv = self.linear1_qkv(x_in).unflatten(-1, (3, self.num_heads, -1))
-        mlp_act = self.linear1_mlp(x_in)  # GELU in the GEMM epilogue
+        qkv_lin = self.linear1_qkv
+        mlp_lin = getattr(self.linear1_mlp, "lin", None)
+        if (hasattr(qkv_lin, "ln_quant") and hasattr(mlp_lin, "mm_fp8")
+                and x.is_cuda and x.dtype == torch.bfloat16
+                and ops.hip_available("layer_norm_mod_fp8")):
+            # fp8 mode: ONE fused LN+quant pass shared by both consumers
+            # (qkv and mlp-up read the same normalized activation) — was
+            # one LN + two standalone quant passes.
+            x8, s_used = qkv_lin.ln_quant(x, mod.scale, mod.shift)
+            qkv = qkv_lin.mm_fp8(x8, s_used)
+            mlp_act = ops.gelu_tanh(mlp_lin.mm_fp8(x8, s_used))
+        else:
+            x_in = ops.layer_norm_mod(x, mod.scale, mod.shift)
+            qkv = self.linear1_qkv(x_in)
+            mlp_act = self.linear1_mlp(x_in)  # GELU in the GEMM epilogue
+        qkv = qkv.unflatten(-1, (3, self.num_heads, -1))
         q, k, v = qkv.unbind(2)  # [B,S,H,D] views
         ops.qk_norm_rope_(
             q, k, self.norm.query_norm.scale, self.norm.key_norm.scale, pe

@@ -93,6 +93,49 @@ class FP8Linear(nn.Module):
         bias = lin.bias.detach().clone() if lin.bias is not None else None
         return cls(w8, w_scale.to(w.device), bias, lin.weight.dtype)
 
+    def _warm_from(self, amax0: torch.Tensor) -> None:
+        self.a_amax[:1].copy_(amax0)
+        self.x_scale.copy_((amax0 / FP8_MAX).clamp(min=1e-12))
+        self._warm = True
+
+    def mm_fp8(self, x8: torch.Tensor, scale_used: torch.Tensor) -> torch.Tensor:
+        """fp8 GEMM on an already-quantized activation (+ its used scale)."""
+        shape = x8.shape
+        y = torch._scaled_mm(
+            x8.reshape(-1, shape[-1]), self.weight_fp8.t(),
+            scale_a=scale_used.reshape(()), scale_b=self.w_scale,
+            bias=self.bias.to(self.out_dtype) if self.bias is not None else None,
+            out_dtype=self.out_dtype,
+        )
+        return y.reshape(*shape[:-1], y.shape[-1])
+
+    def ln_quant(self, x: torch.Tensor, mscale: torch.Tensor,
+                 mshift: torch.Tensor):
+        """AdaLN-modulated LayerNorm with the fp8 cast fused into the LN
+        kernel (no standalone quant pass). Returns (x8 [B,S,D], scale_used).
+        Callers may share x8/scale across several consumers of the same
+        normalized activation (e.g. single-block qkv + mlp-up)."""
+        from .. import ops
+
+        if not self._warm:
+            ln = ops.layer_norm_mod(x, mscale, mshift)
+            self._warm_from(ln.abs().amax().reshape(1).float())
+        x8 = ops.layer_norm_mod_fp8(x, mscale, mshift, self.x_scale,
+                                    self.a_amax, self.x_scale_used)
+        return x8, self.x_scale_used
+
+    def forward_ln(self, x: torch.Tensor, mscale: torch.Tensor,
+                   mshift: torch.Tensor) -> torch.Tensor:
+        """layer_norm_mod -> this Linear, with the quant fused into the LN
+        kernel when available (fp8 serving hot path)."""
+        from .. import ops
+
+        if (x.is_cuda and x.dtype == torch.bfloat16
+                and ops.hip_available("layer_norm_mod_fp8")):
+            x8, s_used = self.ln_quant(x, mscale, mshift)
+            return self.mm_fp8(x8, s_used)
+        return self.forward(ops.layer_norm_mod(x, mscale, mshift))
+
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         from .. import ops
 
@@ -102,10 +145,7 @@ class FP8Linear(nn.Module):
                 and ops.hip_available("quant_fp8")):
             if not self._warm:
                 # first call: measure directly (still async, device-side)
-                amax0 = x2.abs().amax().reshape(1).float()
-                self.a_amax[:1].copy_(amax0)
-                self.x_scale.copy_((amax0 / FP8_MAX).clamp(min=1e-12))
-                self._warm = True
+                self._warm_from(x2.abs().amax().reshape(1).float())
             # the kernel quantizes with x_scale, updates the running amax,
             # writes the NEXT call's x_scale, and snapshots the scale it
             # used into x_scale_used — no host-side scale math

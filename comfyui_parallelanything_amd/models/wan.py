@@ -16,7 +16,7 @@ import torch
 from torch import nn
 
 from .. import ops
-from .layers import FusedMLP, MLPEmbedder, QKNorm
+from .layers import FusedMLP, MLPEmbedder, QKNorm, ln_mod_into
 
 
 def rope_3d_table(f: int, h: int, w: int, axes_dim: Tuple[int, ...],
@@ -54,8 +54,9 @@ class WanBlock(nn.Module):
         m = (e + self.mod).unbind(dim=1)  # 6 x [B, dim]
         shift1, scale1, gate1, shift2, scale2, gate2 = m
 
-        h = ops.layer_norm_mod(x, scale1, shift1)
-        qkv = self.self_qkv(h).unflatten(-1, (3, self.num_heads, -1))
+        qkv = ln_mod_into(self.self_qkv, x, scale1, shift1).unflatten(
+            -1, (3, self.num_heads, -1)
+        )
         q, k, v = qkv.unbind(2)  # [B,S,H,D] views
         ops.qk_norm_rope_(
             q, k, self.self_norm.query_norm.scale,
@@ -73,8 +74,9 @@ class WanBlock(nn.Module):
             ops.attention_bshd(q, k, v, self.scale).flatten(2)
         )
 
-        h = ops.layer_norm_mod(x, scale2, shift2)
-        return ops.gate_residual(x, gate2, self.ffn(h))
+        return ops.gate_residual(
+            x, gate2, self.ffn.forward_ln(x, scale2, shift2)
+        )
 
 
 @dataclass

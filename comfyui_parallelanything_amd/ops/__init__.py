@@ -309,6 +309,19 @@ def timestep_embedding(t, dim: int, max_period: float = 10000.0,
     return reference.timestep_embedding(t, dim, max_period, time_factor)
 
 
+def layer_norm_mod_fp8(x, scale, shift, qscale, amax_buf, scale_used,
+                       eps: float = 1e-6) -> torch.Tensor:
+    """AdaLN-modulated LayerNorm emitting e4m3fn directly (fp8 serving
+    mode): the bf16->fp8 cast fuses into the normalize pass, replacing
+    layer_norm_mod + quant_fp8 (two HBM round-trips -> one). Delayed
+    scaling matches quant_fp8: quantizes with qscale[0], updates the
+    running amax, writes the next scale, snapshots the used scale into
+    scale_used. GPU-only."""
+    ext = _require_ext("layer_norm_mod_fp8")
+    return ext.layer_norm_mod_fp8(x, scale, shift, qscale, amax_buf,
+                                  scale_used, float(eps))
+
+
 def timestep_embed_mlp(t, w1, b1, max_period: float = 10000.0,
                        time_factor: float = 1000.0) -> torch.Tensor:
     """Fused sinusoidal embedding + first Linear + SiLU: [B] -> [B, H].

@@ -265,6 +265,89 @@ __global__ void layer_norm_mod_bf16_kernel(const bf16* __restrict__ x,
     }
 }
 
+
+// AdaLN-modulated LayerNorm with the bf16 -> fp8(e4m3fn) cast fused into
+// the normalize pass (fp8 serving mode): kills the standalone quant_fp8
+// pass per projection (one HBM round-trip of [B,S,D] saved per call).
+// Delayed scaling identical to quant_fp8_bf16_kernel: quantizes with
+// qscale[0], atomically maxes |out| into amax_buf[0], the LAST block
+// (amax_buf[1] = counter) writes the NEXT call's scale and snapshots the
+// used scale into scale_used[0] for _scaled_mm.
+__global__ void layer_norm_mod_fp8_kernel(
+    const bf16* __restrict__ x, const bf16* __restrict__ scale_m,
+    const bf16* __restrict__ shift, unsigned char* __restrict__ out,
+    int S, int D, float eps, float* __restrict__ qscale,
+    float* __restrict__ amax_buf, float* scale_used) {
+    const long row = blockIdx.x;
+    const long b = row / S;
+    const short8* xr = reinterpret_cast<const short8*>(x + row * (long)D);
+    const short8* sc = reinterpret_cast<const short8*>(scale_m + b * (long)D);
+    const short8* sh = reinterpret_cast<const short8*>(shift + b * (long)D);
+    const int DV = D / 8;
+    __shared__ float scratch[8];
+    const float s_entry = qscale[0];
+    const float inv_s = 1.0f / s_entry;
+
+    float s1 = 0.f, s2 = 0.f;
+    for (int i = threadIdx.x; i < DV; i += blockDim.x) {
+        short8 v = xr[i];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            float f = bf2f(__ushort_as_bfloat16((unsigned short)v[j]));
+            s1 += f;
+            s2 += f * f;
+        }
+    }
+    float mean = block_reduce_sum<256>(s1, scratch) / (float)D;
+    float var = block_reduce_sum<256>(s2, scratch) / (float)D - mean * mean;
+    const float rstd = rsqrtf(var + eps);
+    float local_amax = 0.f;
+    for (int i = threadIdx.x; i < DV; i += blockDim.x) {
+        short8 v = xr[i], a = sc[i], c = sh[i];
+        unsigned char pack[8];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            float f = (bf2f(__ushort_as_bfloat16((unsigned short)v[j])) - mean) * rstd;
+            f = f * (1.f + bf2f(__ushort_as_bfloat16((unsigned short)a[j]))) +
+                bf2f(__ushort_as_bfloat16((unsigned short)c[j]));
+            local_amax = fmaxf(local_amax, fabsf(f));
+            float qv = fminf(fmaxf(f * inv_s, -448.f), 448.f);
+            pack[j] = (unsigned char)__hip_cvt_float_to_fp8(
+                qv, __HIP_SATFINITE, __HIP_E4M3);
+        }
+        *reinterpret_cast<unsigned long long*>(&out[row * (long)D + i * 8]) =
+            *reinterpret_cast<unsigned long long*>(pack);
+    }
+    // block amax -> global; scratch reuse needs a barrier after the LN reduce
+    __syncthreads();
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1)
+        local_amax = fmaxf(local_amax, __shfl_xor(local_amax, off, 64));
+    const int lane = threadIdx.x & 63;
+    const int wid = threadIdx.x >> 6;
+    if (lane == 0) scratch[wid] = local_amax;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        float m = 0.f;
+        for (int i = 0; i < (int)(blockDim.x / 64); ++i)
+            m = fmaxf(m, scratch[i]);
+        atomicMax(reinterpret_cast<unsigned int*>(amax_buf),
+                  __float_as_uint(m));
+        __threadfence();
+        const unsigned int done = atomicAdd(
+            reinterpret_cast<unsigned int*>(&amax_buf[1]), 1u);
+        if (done == gridDim.x - 1) {
+            __threadfence();
+            reinterpret_cast<unsigned int*>(amax_buf)[1] = 0u;
+            const float next = amax_buf[0] * 0.999f;
+            amax_buf[0] = next;
+            if (scale_used != qscale) scale_used[0] = s_entry;
+            qscale[0] = fmaxf(next / 448.f, 1e-12f);
+            __threadfence();
+        }
+    }
+}
+
 // bf16 fast path: 8-wide gated residual
 __global__ void gate_residual_bf16_kernel(const bf16* __restrict__ res,
                                           const bf16* __restrict__ gate,
@@ -970,6 +1053,29 @@ at::Tensor layer_norm_mod(at::Tensor x, at::Tensor scale, at::Tensor shift,
     return out;
 }
 
+at::Tensor layer_norm_mod_fp8(at::Tensor x, at::Tensor scale, at::Tensor shift,
+                              at::Tensor qscale, at::Tensor amax_buf,
+                              at::Tensor scale_used, double eps) {
+    CHECK_GPU(x);
+    TORCH_CHECK(x.dim() == 3, "layer_norm_mod_fp8 expects [B, S, D]");
+    TORCH_CHECK(x.scalar_type() == at::kBFloat16, "bf16 input");
+    auto xc = x.contiguous();
+    auto sc = scale.contiguous();
+    auto sh = shift.contiguous();
+    const int S = (int)xc.size(1), D = (int)xc.size(2);
+    TORCH_CHECK((D % 8) == 0, "layer_norm_mod_fp8: D % 8 == 0");
+    TORCH_CHECK(amax_buf.numel() >= 2, "amax_buf needs the counter slot");
+    auto out = at::empty(xc.sizes(), xc.options().dtype(at::kFloat8_e4m3fn));
+    const dim3 grid((unsigned)(xc.size(0) * S));
+    hipLaunchKernelGGL(layer_norm_mod_fp8_kernel, grid, dim3(256), 0,
+                       cur_stream(), (const bf16*)xc.data_ptr(),
+                       (const bf16*)sc.data_ptr(), (const bf16*)sh.data_ptr(),
+                       (unsigned char*)out.data_ptr(), S, D, (float)eps,
+                       qscale.data_ptr<float>(), amax_buf.data_ptr<float>(),
+                       scale_used.data_ptr<float>());
+    return out;
+}
+
 at::Tensor gate_residual(at::Tensor res, at::Tensor gate, at::Tensor x) {
     CHECK_GPU(x);
     TORCH_CHECK(x.dim() == 3 && gate.dim() == 2, "x [B,S,D], gate [B,D]");
@@ -1282,9 +1388,12 @@ __global__ __launch_bounds__(512, 2) void attn_fwd_v4_kernel(
     constexpr int KK = D / 16;        // QK^T K-steps per 32-key tile
     constexpr int NV = D / 32;        // PV dim tiles
     constexpr int KVECS = (KVBLK * D) / (8 * THREADS);
+    // D=128: double-buffered LDS (75.8 KB, still 2 blocks/CU) for the v5a
+    // 1-barrier pipelined schedule; D=64 keeps the v4 2-barrier loop.
+    constexpr int DBUF = (D == 128) ? 2 : 1;
 
-    __shared__ bf16 k_lds[KVBLK * KPAD];
-    __shared__ bf16 v_lds[KVBLK * VROW];
+    __shared__ bf16 k_lds[DBUF * KVBLK * KPAD];
+    __shared__ bf16 v_lds[DBUF * KVBLK * VROW];
 
     const int tid = threadIdx.x;
     const int lane = tid & 63;
@@ -1358,51 +1467,54 @@ __global__ __launch_bounds__(512, 2) void attn_fwd_v4_kernel(
             }
         }
     };
-    auto write_tile_lds = [&]() {
+    auto write_k_lds = [&](int buf) {
 #pragma unroll
         for (int i = 0; i < KVECS; ++i) {
             const int idx = tid + i * THREADS;
             const int row = idx / (D / 8);
             const int col = (idx % (D / 8)) * 8;
-            *reinterpret_cast<bf16x8*>(&k_lds[row * KPAD + col]) = kreg[i];
+            *reinterpret_cast<bf16x8*>(
+                &k_lds[buf * (KVBLK * KPAD) + row * KPAD + col]) = kreg[i];
+        }
+    };
+    auto write_v_lds = [&](int buf) {
+#pragma unroll
+        for (int i = 0; i < KVECS; ++i) {
+            const int idx = tid + i * THREADS;
+            const int row = idx / (D / 8);
+            const int col = (idx % (D / 8)) * 8;
             // row-major V, b128 store; column XOR by (row&8)<<1 keeps the
             // tr gather conflict-free under either half-wave pairing
             *reinterpret_cast<bf16x8*>(
-                &v_lds[row * VROW + (col ^ ((row & 8) << 1))]) = vreg[i];
+                &v_lds[buf * (KVBLK * VROW) + row * VROW +
+                       (col ^ ((row & 8) << 1))]) = vreg[i];
         }
     };
 
-    issue_tile_loads(0);
-    const int n_tiles = (Sk + KVBLK - 1) / KVBLK;
-    for (int t = 0; t < n_tiles; ++t) {
-        const int kv0 = t * KVBLK;
-        __syncthreads();
-        write_tile_lds();
-        __syncthreads();
-        if (t + 1 < n_tiles) issue_tile_loads(kv0 + KVBLK);
-
-        // ---- swapped QK^T: S^T[key][row] for two 32-key tiles ------------
-        f32x16 st[2];
+    // ---- per-tile phases --------------------------------------------------
+    auto qk_phase = [&](int buf, f32x16* st) {
+        // swapped QK^T: S^T[key][row] for two 32-key tiles;
+        // A = K chunk: lane holds K[kt*32 + l32][kk*16 + hi*8 + j]
+        __builtin_amdgcn_s_setprio(1);
 #pragma unroll
         for (int kt = 0; kt < 2; ++kt) {
 #pragma unroll
             for (int r = 0; r < 16; ++r) st[kt][r] = 0.f;
-            __builtin_amdgcn_s_setprio(1);
 #pragma unroll
             for (int kk = 0; kk < KK; ++kk) {
-                // A = K chunk: lane holds K[kt*32 + l32][kk*16 + hi*8 + j]
                 bf16x8 afrag = *reinterpret_cast<const bf16x8*>(
-                    &k_lds[(kt * 32 + l32) * KPAD + kk * 16 + hi * 8]);
+                    &k_lds[buf * (KVBLK * KPAD) + (kt * 32 + l32) * KPAD +
+                           kk * 16 + hi * 8]);
                 st[kt] = mfma32x32x16(afrag, qfrag[kk], st[kt]);
             }
-            __builtin_amdgcn_s_setprio(0);
         }
-
-        // ---- lane-local online softmax (this lane's row = q0 + l32) ------
+        __builtin_amdgcn_s_setprio(0);
+    };
+    auto softmax_phase = [&](int kv0, f32x16* st, bf16x8* pfrag) {
+        // lane-local online softmax (this lane's row = q0 + l32);
         // value (kt, reg) = S[row][key = kt*32 + (reg&3) + 8*(reg>>2) + 4*hi]
-        // track the RAW tile max (scale2 > 0 commutes with max) and fuse
-        // the softmax scale into the exp2 argument as one FMA per element —
-        // saves a 32-wide VALU multiply pass per tile.
+        // raw tile max (scale2 > 0 commutes with max), softmax scale fused
+        // into the exp2 argument as one FMA per element.
         float mx = -3e30f;
 #pragma unroll
         for (int kt = 0; kt < 2; ++kt)
@@ -1413,7 +1525,7 @@ __global__ __launch_bounds__(512, 2) void attn_fwd_v4_kernel(
                 st[kt][r] = sv;
                 mx = fmaxf(mx, sv);
             }
-        mx = fmaxf(mx, __shfl_xor(mx, 32, 64));  // partner holds the row's other 32 keys
+        mx = fmaxf(mx, __shfl_xor(mx, 32, 64));  // partner: row's other 32 keys
         const float mnew = fmaxf(m_run, mx * scale2);
         const float alpha = __builtin_amdgcn_exp2f(m_run - mnew);
         m_run = mnew;
@@ -1435,11 +1547,9 @@ __global__ __launch_bounds__(512, 2) void attn_fwd_v4_kernel(
 #pragma unroll
                 for (int r = 0; r < 16; ++r) o_acc[n][r] *= alpha;
         }
-
-        // ---- P f32 -> bf16 fragments via cvt_pk + permlane32_swap --------
-        // chunk c (16 keys) uses regs 8*(c&1)..8*(c&1)+7 of st[c>>1]; after
-        // the half-swap each lane holds P[row l32][chunk base + hi*8 + j].
-        bf16x8 pfrag[4];
+        // P f32 -> bf16 fragments via cvt_pk + permlane32_swap: chunk c
+        // (16 keys) uses regs 8*(c&1)..+7 of st[c>>1]; after the half-swap
+        // each lane holds P[row l32][chunk base + hi*8 + j].
 #pragma unroll
         for (int c = 0; c < 4; ++c) {
             const f32x16& sv = st[c >> 1];
@@ -1454,11 +1564,10 @@ __global__ __launch_bounds__(512, 2) void attn_fwd_v4_kernel(
                                  (unsigned int)r02[1], (unsigned int)r13[1]};
             pfrag[c] = *reinterpret_cast<bf16x8*>(d);
         }
-
-        // ---- PV: O^T[dim][row] += V^T chunk · P^T chunk ------------------
-        // A = V^T: lane holds V^T[n*32 + l32][16c + hi*8 + j], gathered
-        // from the row-major V image by ds_read_b64_tr_b16 (2 reads per
-        // (c, n): keys +0..3 and +4..7), base VGPR + immediate offsets.
+    };
+    auto pv_phase = [&](int buf, bf16x8* pfrag) {
+        // PV: O^T[dim][row] += V^T chunk · P^T chunk; A = V^T gathered from
+        // the row-major V image by ds_read_b64_tr_b16 (2 reads per (c, n)).
         __builtin_amdgcn_s_setprio(1);
 #pragma unroll
         for (int c = 0; c < 4; ++c) {
@@ -1466,16 +1575,60 @@ __global__ __launch_bounds__(512, 2) void attn_fwd_v4_kernel(
             for (int n = 0; n < NV; ++n) {
                 short4v alo = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
                     (__attribute__((address_space(3))) short4v*)
-                        &v_lds[trb + c * (16 * VROW) + n * 32]);
+                        &v_lds[buf * (KVBLK * VROW) + trb + c * (16 * VROW) +
+                               n * 32]);
                 short4v ahi = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
                     (__attribute__((address_space(3))) short4v*)
-                        &v_lds[trb + c * (16 * VROW) + 4 * VROW + n * 32]);
+                        &v_lds[buf * (KVBLK * VROW) + trb + c * (16 * VROW) +
+                               4 * VROW + n * 32]);
                 bf16x8 va = __builtin_shufflevector(alo, ahi,
                                                     0, 1, 2, 3, 4, 5, 6, 7);
                 o_acc[n] = mfma32x32x16(va, pfrag[c], o_acc[n]);
             }
         }
         __builtin_amdgcn_s_setprio(0);
+    };
+
+    const int n_tiles = (Sk + KVBLK - 1) / KVBLK;
+    if constexpr (DBUF == 2) {
+        // v5a pipelined schedule (measured +4.0% flux / +4.2% long-S over
+        // the 2-barrier loop, scripts/attn_v5.hip): double-buffered LDS,
+        // ONE barrier per tile; tile t+1's stores land in the other buffer
+        // interleaved where the LDS port is idle (K after QK^T, V after
+        // softmax); tile t+2's global loads issue before PV.
+        issue_tile_loads(0);
+        write_k_lds(0);
+        write_v_lds(0);
+        if (n_tiles > 1) issue_tile_loads(KVBLK);
+        __syncthreads();
+        for (int t = 0; t < n_tiles; ++t) {
+            const int p = t & 1;
+            f32x16 st[2];
+            bf16x8 pfrag[4];
+            qk_phase(p, st);
+            if (t + 1 < n_tiles) write_k_lds(p ^ 1);
+            softmax_phase(t * KVBLK, st, pfrag);
+            if (t + 1 < n_tiles) write_v_lds(p ^ 1);
+            if (t + 2 < n_tiles) issue_tile_loads((t + 2) * KVBLK);
+            pv_phase(p, pfrag);
+            __syncthreads();
+        }
+    } else {
+        // D=64 keeps the v4 2-barrier loop (v5a measured -2.6% there —
+        // halved MFMA per tile leaves too little compute to hide stores).
+        issue_tile_loads(0);
+        for (int t = 0; t < n_tiles; ++t) {
+            __syncthreads();
+            write_k_lds(0);
+            write_v_lds(0);
+            __syncthreads();
+            if (t + 1 < n_tiles) issue_tile_loads((t + 1) * KVBLK);
+            f32x16 st[2];
+            bf16x8 pfrag[4];
+            qk_phase(0, st);
+            softmax_phase(t * KVBLK, st, pfrag);
+            pv_phase(0, pfrag);
+        }
     }
 
     // ---- epilogue: O = O^T / l, row is lane-local -------------------------
@@ -1746,6 +1899,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("rms_norm", &rms_norm, "RMSNorm (gfx950)",
           py::arg("x"), py::arg("weight") = py::none(), py::arg("eps") = 1e-6);
     m.def("layer_norm_mod", &layer_norm_mod, "AdaLN-modulated LayerNorm (gfx950)");
+    m.def("layer_norm_mod_fp8", &layer_norm_mod_fp8,
+          "AdaLN LayerNorm with fused e4m3fn quant + delayed scaling (gfx950)");
     m.def("gate_residual", &gate_residual, "Gated residual add (gfx950)");
     m.def("group_norm_silu", &group_norm_silu, "GroupNorm+SiLU (gfx950)",
           py::arg("x"), py::arg("groups"), py::arg("weight") = py::none(),

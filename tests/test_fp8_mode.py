@@ -151,3 +151,84 @@ def test_fp8_linear_amax_jump_uses_snapshot_scale():
     ref = lin(x2).float()
     rel = (out - ref).norm() / ref.norm()
     assert rel < 0.10, f"fp8 amax-jump error {rel:.4f}"
+
+
+def test_layer_norm_mod_fp8_fused():
+    """Fused AdaLN+fp8 kernel == layer_norm_mod -> quantize, and maintains
+    the same delayed-scaling state contract as quant_fp8."""
+    from comfyui_parallelanything_amd import ops
+
+    if not ops.hip_available("layer_norm_mod_fp8"):
+        pytest.skip("no layer_norm_mod_fp8 in extension")
+    torch.manual_seed(5)
+    B, S, D = 2, 64, 1024
+    x = torch.randn(B, S, D, device="cuda", dtype=torch.bfloat16)
+    sc = torch.randn(B, D, device="cuda", dtype=torch.bfloat16) * 0.1
+    sh = torch.randn(B, D, device="cuda", dtype=torch.bfloat16) * 0.1
+    ln = ops.layer_norm_mod(x, sc, sh)
+    true_amax = ln.float().abs().amax().item()
+    s0 = true_amax / 448.0
+    qscale = torch.tensor([s0], device="cuda")
+    amax = torch.zeros(2, device="cuda")
+    used = torch.zeros(1, device="cuda")
+    x8 = ops.layer_norm_mod_fp8(x, sc, sh, qscale, amax, used)
+    assert x8.dtype == torch.float8_e4m3fn and x8.shape == x.shape
+    deq = x8.float() * used
+    rel = (deq - ln.float()).norm() / ln.float().norm()
+    assert rel < 0.05, f"fused LN+quant error {rel:.4f}"
+    # delayed-scaling contract: amax tracked, counter reset, next scale
+    assert abs(amax[0].item() - true_amax * 0.999) < 2e-2
+    assert amax[1].item() == 0.0
+    assert used.item() == qscale.new_tensor([s0]).item()
+
+
+def test_fp8_linear_forward_ln_matches_composed():
+    """FP8Linear.forward_ln (fused LN+quant) ~= LN then FP8Linear."""
+    from comfyui_parallelanything_amd import ops
+    from comfyui_parallelanything_amd.models.quant import (
+        FP8Linear, _supports_scaled_mm,
+    )
+
+    if not _supports_scaled_mm() or not ops.hip_available("layer_norm_mod_fp8"):
+        pytest.skip("no fp8 fused-LN path")
+    torch.manual_seed(6)
+    lin = torch.nn.Linear(1024, 1024).cuda().to(torch.bfloat16)
+    q1 = FP8Linear.from_linear(lin)
+    q2 = FP8Linear.from_linear(lin)
+    x = torch.randn(2, 64, 1024, device="cuda", dtype=torch.bfloat16)
+    sc = torch.randn(2, 1024, device="cuda", dtype=torch.bfloat16) * 0.1
+    sh = torch.randn(2, 1024, device="cuda", dtype=torch.bfloat16) * 0.1
+    fused = q1.forward_ln(x, sc, sh).float()
+    composed = q2(ops.layer_norm_mod(x, sc, sh)).float()
+    rel = (fused - composed).norm() / composed.norm()
+    assert rel < 0.02, f"fused vs composed {rel:.4f}"
+
+
+def test_fp8_model_uses_fused_ln(monkeypatch):
+    """quantized flux-tiny routes through ln_quant (no standalone quant
+    pass on the block LN sites) and stays finite."""
+    from comfyui_parallelanything_amd import ops
+    from comfyui_parallelanything_amd.models.quant import (
+        FP8Linear, _supports_scaled_mm, quantize_fp8,
+    )
+    from comfyui_parallelanything_amd.models.registry import (
+        flux_inputs, make_flux,
+    )
+
+    if not _supports_scaled_mm() or not ops.hip_available("layer_norm_mod_fp8"):
+        pytest.skip("no fp8 fused-LN path")
+    m = make_flux(dev="cuda", dtype=torch.bfloat16, tiny=True)
+    quantize_fp8(m, min_features=32)
+    calls = {"n": 0}
+    orig = FP8Linear.ln_quant
+
+    def spy(self, *a, **k):
+        calls["n"] += 1
+        return orig(self, *a, **k)
+
+    monkeypatch.setattr(FP8Linear, "ln_quant", spy)
+    x, t, c, kw = flux_inputs(2, dev="cuda", dtype=torch.bfloat16, tiny=True)
+    with torch.no_grad():
+        out = m(x, t, context=c, **kw)
+    assert torch.isfinite(out.float()).all()
+    assert calls["n"] > 0, "fused LN+quant path not taken"
