@@ -113,8 +113,9 @@ class MLPEmbedder(nn.Module):
         fusion; one launch instead of three and the sinusoid stays fp32
         into the accumulate), then the out_layer GEMM. CPU / non-bf16:
         composed ops."""
-        w = self.in_layer.weight
-        if (t.is_cuda and w.dtype == torch.bfloat16
+        # in_layer may be FP8Linear-swapped (no .weight attr): compose then
+        w = getattr(self.in_layer, "weight", None)
+        if (w is not None and t.is_cuda and w.dtype == torch.bfloat16
                 and w.shape[1] % 8 == 0
                 and ops.hip_available("timestep_embed_mlp")):
             h = ops.timestep_embed_mlp(t, w, self.in_layer.bias,
@@ -122,7 +123,9 @@ class MLPEmbedder(nn.Module):
             return self.out_layer(h)
         emb = ops.timestep_embedding(t, self.in_layer.in_features,
                                      max_period, time_factor)
-        return self.forward(emb.to(w.dtype))
+        return self.forward(
+            emb.to(w.dtype if w is not None else torch.bfloat16)
+        )
 
 
 class RMSNorm(nn.Module):

@@ -276,49 +276,58 @@ __global__ void layer_norm_mod_bf16_kernel(const bf16* __restrict__ x,
 __global__ void layer_norm_mod_fp8_kernel(
     const bf16* __restrict__ x, const bf16* __restrict__ scale_m,
     const bf16* __restrict__ shift, unsigned char* __restrict__ out,
-    int S, int D, float eps, float* __restrict__ qscale,
+    long n_rows, int S, int D, float eps, float* __restrict__ qscale,
     float* __restrict__ amax_buf, float* scale_used) {
-    const long row = blockIdx.x;
-    const long b = row / S;
-    const short8* xr = reinterpret_cast<const short8*>(x + row * (long)D);
-    const short8* sc = reinterpret_cast<const short8*>(scale_m + b * (long)D);
-    const short8* sh = reinterpret_cast<const short8*>(shift + b * (long)D);
     const int DV = D / 8;
     __shared__ float scratch[8];
     const float s_entry = qscale[0];
     const float inv_s = 1.0f / s_entry;
 
-    float s1 = 0.f, s2 = 0.f;
-    for (int i = threadIdx.x; i < DV; i += blockDim.x) {
-        short8 v = xr[i];
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-            float f = bf2f(__ushort_as_bfloat16((unsigned short)v[j]));
-            s1 += f;
-            s2 += f * f;
-        }
-    }
-    float mean = block_reduce_sum<256>(s1, scratch) / (float)D;
-    float var = block_reduce_sum<256>(s2, scratch) / (float)D - mean * mean;
-    const float rstd = rsqrtf(var + eps);
+    // Grid-stride over rows with ONE amax atomic + counter hit per BLOCK
+    // (a block-per-row launch would serialize tens of thousands of atomics
+    // on amax_buf — measured as a ~100 ms/step regression on flux fp8).
     float local_amax = 0.f;
-    for (int i = threadIdx.x; i < DV; i += blockDim.x) {
-        short8 v = xr[i], a = sc[i], c = sh[i];
-        unsigned char pack[8];
+    for (long row = blockIdx.x; row < n_rows; row += gridDim.x) {
+        const long b = row / S;
+        const short8* xr = reinterpret_cast<const short8*>(x + row * (long)D);
+        const short8* sc =
+            reinterpret_cast<const short8*>(scale_m + b * (long)D);
+        const short8* sh =
+            reinterpret_cast<const short8*>(shift + b * (long)D);
+
+        float s1 = 0.f, s2 = 0.f;
+        for (int i = threadIdx.x; i < DV; i += blockDim.x) {
+            short8 v = xr[i];
 #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-            float f = (bf2f(__ushort_as_bfloat16((unsigned short)v[j])) - mean) * rstd;
-            f = f * (1.f + bf2f(__ushort_as_bfloat16((unsigned short)a[j]))) +
-                bf2f(__ushort_as_bfloat16((unsigned short)c[j]));
-            local_amax = fmaxf(local_amax, fabsf(f));
-            float qv = fminf(fmaxf(f * inv_s, -448.f), 448.f);
-            pack[j] = (unsigned char)__hip_cvt_float_to_fp8(
-                qv, __HIP_SATFINITE, __HIP_E4M3);
+            for (int j = 0; j < 8; ++j) {
+                float f = bf2f(__ushort_as_bfloat16((unsigned short)v[j]));
+                s1 += f;
+                s2 += f * f;
+            }
         }
-        *reinterpret_cast<unsigned long long*>(&out[row * (long)D + i * 8]) =
-            *reinterpret_cast<unsigned long long*>(pack);
+        float mean = block_reduce_sum<256>(s1, scratch) / (float)D;
+        float var = block_reduce_sum<256>(s2, scratch) / (float)D - mean * mean;
+        const float rstd = rsqrtf(var + eps);
+        for (int i = threadIdx.x; i < DV; i += blockDim.x) {
+            short8 v = xr[i], a = sc[i], c = sh[i];
+            unsigned char pack[8];
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                float f = (bf2f(__ushort_as_bfloat16((unsigned short)v[j])) - mean) * rstd;
+                f = f * (1.f + bf2f(__ushort_as_bfloat16((unsigned short)a[j]))) +
+                    bf2f(__ushort_as_bfloat16((unsigned short)c[j]));
+                local_amax = fmaxf(local_amax, fabsf(f));
+                float qv = fminf(fmaxf(f * inv_s, -448.f), 448.f);
+                pack[j] = (unsigned char)__hip_cvt_float_to_fp8(
+                    qv, __HIP_SATFINITE, __HIP_E4M3);
+            }
+            *reinterpret_cast<unsigned long long*>(
+                &out[row * (long)D + i * 8]) =
+                *reinterpret_cast<unsigned long long*>(pack);
+        }
     }
-    // block amax -> global; scratch reuse needs a barrier after the LN reduce
+    // protect scratch against stragglers still reading the last row's
+    // block_reduce_sum broadcast
     __syncthreads();
 #pragma unroll
     for (int off = 32; off > 0; off >>= 1)
@@ -1066,12 +1075,16 @@ at::Tensor layer_norm_mod_fp8(at::Tensor x, at::Tensor scale, at::Tensor shift,
     TORCH_CHECK((D % 8) == 0, "layer_norm_mod_fp8: D % 8 == 0");
     TORCH_CHECK(amax_buf.numel() >= 2, "amax_buf needs the counter slot");
     auto out = at::empty(xc.sizes(), xc.options().dtype(at::kFloat8_e4m3fn));
-    const dim3 grid((unsigned)(xc.size(0) * S));
+    const long n_rows = xc.size(0) * (long)S;
+    // cap the grid: the delayed-scaling epilogue costs one global atomic
+    // pair per BLOCK (a block-per-row grid serialized ~37k atomics/call)
+    const dim3 grid((unsigned)std::min<long>(n_rows, 2048));
     hipLaunchKernelGGL(layer_norm_mod_fp8_kernel, grid, dim3(256), 0,
                        cur_stream(), (const bf16*)xc.data_ptr(),
                        (const bf16*)sc.data_ptr(), (const bf16*)sh.data_ptr(),
-                       (unsigned char*)out.data_ptr(), S, D, (float)eps,
-                       qscale.data_ptr<float>(), amax_buf.data_ptr<float>(),
+                       (unsigned char*)out.data_ptr(), n_rows, S, D,
+                       (float)eps, qscale.data_ptr<float>(),
+                       amax_buf.data_ptr<float>(),
                        scale_used.data_ptr<float>());
     return out;
 }
