@@ -55,6 +55,14 @@ PA_DEV unsigned int cvt_pk_bf16(float lo, float hi) {
 //          e4m3 with per-tensor scales folded into the softmax scale;
 //          the A/B byte->k map is freely chosen — it cancels as long as
 //          BOTH operands use the same packing, verified by mx_probe)
+//      4 = 4-slot LDS ring, ONE barrier every TWO tiles (writes land two
+//          tiles ahead; wave skew bounded by the ring depth). LDS 151.6 KB
+//          -> 1 block/CU: tests whether intra-block desync beats the
+//          co-resident second block.
+//      5 = cross-tile stagger: iter t runs QK^T(t) then PV(t-1) back to
+//          back (matrix beside matrix), softmax(t)+stores in the second
+//          barrier segment — barrier pairs the MFMA batches against the
+//          partner wave's memory segment (MI355X_MICROARCH item 5 shape).
 template <int D, int VAR>
 __global__ __launch_bounds__(512, 2) void attn_v5_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ k,
@@ -68,7 +76,7 @@ __global__ __launch_bounds__(512, 2) void attn_v5_kernel(
     constexpr int KK = D / 16;
     constexpr int NV = D / 32;
     constexpr int KVECS = (KVBLK * D) / (8 * THREADS);
-    constexpr int DBUF = (VAR == 0) ? 1 : 2;
+    constexpr int DBUF = (VAR == 0) ? 1 : (VAR == 4 ? 4 : 2);
     constexpr bool FP8QK = (VAR == 3);
     constexpr int K8ROW = D + 16;      // fp8 K image row stride (16B-aligned
                                        // b128 reads, conflict-free 36-dw rows)
@@ -145,8 +153,12 @@ __global__ __launch_bounds__(512, 2) void attn_v5_kernel(
     // (squared) folds into the softmax scale — zero extra VALU per element.
     const float scale2 = (FP8QK ? qk_s2 : 1.f) * scale * PA_LOG2E;
 
-    bf16x8 kreg[KVECS], vreg[KVECS];
-    auto issue_tile_loads = [&](int kv0) {
+    // var4 keeps TWO tiles in flight in registers (slot = tile % 2)
+    constexpr int RSLOTS = (VAR == 4) ? 2 : 1;
+    bf16x8 kreg_s[RSLOTS][KVECS], vreg_s[RSLOTS][KVECS];
+#define kreg kreg_s[0]
+#define vreg vreg_s[0]
+    auto issue_tile_loads_slot = [&](int kv0, int slot) {
 #pragma unroll
         for (int i = 0; i < KVECS; ++i) {
             const int idx = tid + i * THREADS;
@@ -154,17 +166,18 @@ __global__ __launch_bounds__(512, 2) void attn_v5_kernel(
             const int col = (idx % (D / 8)) * 8;
             const int src = kv0 + row;
             if (src < Sk) {
-                kreg[i] = *reinterpret_cast<const bf16x8*>(
+                kreg_s[slot][i] = *reinterpret_cast<const bf16x8*>(
                     kp + (long)src * ss + col);
-                vreg[i] = *reinterpret_cast<const bf16x8*>(
+                vreg_s[slot][i] = *reinterpret_cast<const bf16x8*>(
                     vp + (long)src * ss + col);
             } else {
-                kreg[i] = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
-                vreg[i] = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+                kreg_s[slot][i] = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+                vreg_s[slot][i] = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
             }
         }
     };
-    auto write_k_lds = [&](int buf) {
+    auto issue_tile_loads = [&](int kv0) { issue_tile_loads_slot(kv0, 0); };
+    auto write_k_slot = [&](int buf, int slot) {
 #pragma unroll
         for (int i = 0; i < KVECS; ++i) {
             const int idx = tid + i * THREADS;
@@ -176,7 +189,8 @@ __global__ __launch_bounds__(512, 2) void attn_v5_kernel(
 #pragma unroll
                 for (int j = 0; j < 8; ++j) {
                     float f = bf2f(__ushort_as_bfloat16(
-                                  (unsigned short)kreg[i][j])) * qk_inv_s;
+                                  (unsigned short)kreg_s[slot][i][j])) *
+                              qk_inv_s;
                     pack[j] = (unsigned char)__hip_cvt_float_to_fp8(
                         f, __HIP_SATFINITE, __HIP_E4M3);
                 }
@@ -185,11 +199,12 @@ __global__ __launch_bounds__(512, 2) void attn_v5_kernel(
                     *reinterpret_cast<unsigned long long*>(pack);
             } else {
                 *reinterpret_cast<bf16x8*>(
-                    &k_lds[buf * (KVBLK * KPAD) + row * KPAD + col]) = kreg[i];
+                    &k_lds[buf * (KVBLK * KPAD) + row * KPAD + col]) =
+                    kreg_s[slot][i];
             }
         }
     };
-    auto write_v_lds = [&](int buf) {
+    auto write_v_slot = [&](int buf, int slot) {
 #pragma unroll
         for (int i = 0; i < KVECS; ++i) {
             const int idx = tid + i * THREADS;
@@ -197,9 +212,11 @@ __global__ __launch_bounds__(512, 2) void attn_v5_kernel(
             const int col = (idx % (D / 8)) * 8;
             *reinterpret_cast<bf16x8*>(
                 &v_lds[buf * (KVBLK * VROW) + row * VROW +
-                       (col ^ ((row & 8) << 1))]) = vreg[i];
+                       (col ^ ((row & 8) << 1))]) = vreg_s[slot][i];
         }
     };
+    auto write_k_lds = [&](int buf) { write_k_slot(buf, 0); };
+    auto write_v_lds = [&](int buf) { write_v_slot(buf, 0); };
 
     auto qk_half = [&](int buf, f32x16* st) {
         __builtin_amdgcn_s_setprio(1);
@@ -302,7 +319,61 @@ __global__ __launch_bounds__(512, 2) void attn_v5_kernel(
     };
 
     const int n_tiles = (Sk + KVBLK - 1) / KVBLK;
-    if (VAR == 0) {
+    if constexpr (VAR == 4) {
+        // 4-slot ring, barrier every SECOND tile. Writes land TWO tiles
+        // ahead (tile t+2 from reg slot t%2) so every read is separated
+        // from its write by at least one barrier; the 4-deep ring bounds
+        // WAR hazards across the skew the sparser barriers allow.
+        issue_tile_loads_slot(0, 0);
+        write_k_slot(0, 0);
+        write_v_slot(0, 0);
+        if (n_tiles > 1) {
+            issue_tile_loads_slot(KVBLK, 1);
+            write_k_slot(1, 1);
+            write_v_slot(1, 1);
+        }
+        if (n_tiles > 2) issue_tile_loads_slot(2 * KVBLK, 0);
+        __syncthreads();
+        for (int t = 0; t < n_tiles; ++t) {
+            const int p = t & 3;
+            f32x16 st[2];
+            bf16x8 pfrag[4];
+            qk_half(p, st);
+            if (t + 2 < n_tiles) write_k_slot((t + 2) & 3, t & 1);
+            softmax_half(t * KVBLK, st, pfrag);
+            if (t + 2 < n_tiles) write_v_slot((t + 2) & 3, t & 1);
+            if (t + 3 < n_tiles) issue_tile_loads_slot((t + 3) * KVBLK,
+                                                       (t + 1) & 1);
+            pv_half(p, pfrag);
+            if (t & 1) __syncthreads();
+        }
+    } else if constexpr (VAR == 5) {
+        // cross-tile stagger: [QK^T(t) ; PV(t-1)] in one barrier segment
+        // (matrix beside matrix), [softmax(t) ; stores ; loads] in the
+        // other. o_acc rescale order stays correct because softmax(t)
+        // runs AFTER PV(t-1) lands its m(t-1)-unit contributions.
+        issue_tile_loads(0);
+        write_k_lds(0);
+        write_v_lds(0);
+        if (n_tiles > 1) issue_tile_loads(KVBLK);
+        __syncthreads();
+        bf16x8 pfrag[4];
+        f32x16 st[2];
+        for (int t = 0; t < n_tiles; ++t) {
+            const int p = t & 1;
+            qk_half(p, st);
+            if (t > 0) pv_half(p ^ 1, pfrag);
+            __syncthreads();  // v[p^1] free for the (t+1) stores below
+            softmax_half(t * KVBLK, st, pfrag);
+            if (t + 1 < n_tiles) {
+                write_k_lds(p ^ 1);
+                write_v_lds(p ^ 1);
+            }
+            if (t + 2 < n_tiles) issue_tile_loads((t + 2) * KVBLK);
+            __syncthreads();  // k/v[p^1] visible before QK^T(t+1)
+        }
+        pv_half((n_tiles - 1) & 1, pfrag);  // drain the last tile's PV
+    } else if (VAR == 0) {
         issue_tile_loads(0);
         for (int t = 0; t < n_tiles; ++t) {
             __syncthreads();
@@ -412,13 +483,21 @@ static void launch(int var, const bf16* q, const bf16* k, const bf16* v,
         hipLaunchKernelGGL((attn_v5_kernel<D, 2>), grid, blk, 0, 0, q, k, v,
                            o, S, S, scale, H, inv_s, s2);
         break;
-    default:
+    case 3:
         hipLaunchKernelGGL((attn_v5_kernel<D, 3>), grid, blk, 0, 0, q, k, v,
+                           o, S, S, scale, H, inv_s, s2);
+        break;
+    case 4:
+        hipLaunchKernelGGL((attn_v5_kernel<D, 4>), grid, blk, 0, 0, q, k, v,
+                           o, S, S, scale, H, inv_s, s2);
+        break;
+    default:
+        hipLaunchKernelGGL((attn_v5_kernel<D, 5>), grid, blk, 0, 0, q, k, v,
                            o, S, S, scale, H, inv_s, s2);
     }
 }
 
-constexpr int NVAR = 4;
+constexpr int NVAR = 6;
 
 template <int D>
 static int check_correct(int B, int H, int S, bool spike) {
