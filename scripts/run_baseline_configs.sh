@@ -54,7 +54,8 @@ fi
 run_bench config4_flux 8 --model flux --batch 8 --px 1024
 
 # 5. WAN2.2 I2V 720p batch=4 on 4x MI355X (video / temporal-attn path)
-run_bench config5_wan_i2v 4 --model wan_i2v --batch 4
+# (~45 s/step on ONE GPU at this 4x75600-token config: fewer steps)
+STEPS=2 WARMUP=1 run_bench config5_wan_i2v 4 --model wan_i2v --batch 4
 
 echo "### artifacts:"
 ls -l "$OUT_DIR"
