@@ -441,6 +441,259 @@ __global__ __launch_bounds__(512, 2) void attn_v5_kernel(
     }
 }
 
+#undef kreg
+#undef vreg
+
+// ---------------------------------------------------------------------------
+// var6 — D=64 two-q-tile wave (NOTES r02 item 7): each wave owns 64 q-rows
+// (two 32-row frames). K A-fragments and V tr-reads are shared between the
+// two frames, halving LDS traffic per output row and amortizing the
+// per-tile barrier/stage costs that double D=64's softmax fraction.
+// v5a schedule (double-buffered LDS, one barrier per tile).
+// ---------------------------------------------------------------------------
+template <int D>
+__global__ __launch_bounds__(512, 2) void attn_d64x2_kernel(
+    const bf16* __restrict__ q, const bf16* __restrict__ k,
+    const bf16* __restrict__ v, bf16* __restrict__ out,
+    int S, int Sk, float scale, int H) {
+    constexpr int KVBLK = 64;
+    constexpr int WAVES = 8;
+    constexpr int THREADS = WAVES * 64;
+    constexpr int KPAD = D + 8;
+    constexpr int VROW = 160;
+    constexpr int KK = D / 16;
+    constexpr int NV = D / 32;
+    constexpr int KVECS = (KVBLK * D) / (8 * THREADS);
+    constexpr int QT = 2;
+
+    __shared__ bf16 k_lds[2 * KVBLK * KPAD];
+    __shared__ bf16 v_lds[2 * KVBLK * VROW];
+
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wid = tid >> 6;
+    const int l32 = lane & 31;
+    const int hi = lane >> 5;
+    const int trb = (hi * 8 + ((lane & 15) >> 2)) * VROW +
+                    16 * (((lane >> 4) & 1) ^ hi) + 4 * (lane & 3);
+
+    const int rows_per_blk = WAVES * 32 * QT;  // 512
+    const int nq = (S + rows_per_blk - 1) / rows_per_blk;
+    const long id = blockIdx.x;
+    const long bh = (id & 7) + 8 * ((id >> 3) / nq);
+    const int qtile = (int)((id >> 3) % nq);
+    const long b = bh / H;
+    const int h = (int)(bh % H);
+    const int q0 = qtile * rows_per_blk + wid * (32 * QT);
+    const int ss = H * D;
+
+    const bf16* qp = q + (b * (long)S) * ss + (long)h * D;
+    const bf16* kp = k + (b * (long)Sk) * ss + (long)h * D;
+    const bf16* vp = v + (b * (long)Sk) * ss + (long)h * D;
+    bf16* op = out + (b * (long)S) * ss + (long)h * D;
+
+    bf16x8 qfrag[QT][KK];
+#pragma unroll
+    for (int qt = 0; qt < QT; ++qt) {
+        const int row = q0 + qt * 32 + l32;
+        const int rr = row < S ? row : S - 1;
+#pragma unroll
+        for (int kk = 0; kk < KK; ++kk)
+            qfrag[qt][kk] = *reinterpret_cast<const bf16x8*>(
+                qp + (long)rr * ss + kk * 16 + hi * 8);
+    }
+
+    f32x16 o_acc[QT][NV];
+#pragma unroll
+    for (int qt = 0; qt < QT; ++qt)
+#pragma unroll
+        for (int n = 0; n < NV; ++n)
+#pragma unroll
+            for (int r = 0; r < 16; ++r) o_acc[qt][n][r] = 0.f;
+    float m_run[QT] = {-1e30f, -1e30f}, l_run[QT] = {0.f, 0.f};
+    const float scale2 = scale * PA_LOG2E;
+
+    bf16x8 kreg[KVECS], vreg[KVECS];
+    auto issue_loads = [&](int kv0) {
+#pragma unroll
+        for (int i = 0; i < KVECS; ++i) {
+            const int idx = tid + i * THREADS;
+            const int row = idx / (D / 8);
+            const int col = (idx % (D / 8)) * 8;
+            const int src = kv0 + row;
+            if (src < Sk) {
+                kreg[i] = *reinterpret_cast<const bf16x8*>(
+                    kp + (long)src * ss + col);
+                vreg[i] = *reinterpret_cast<const bf16x8*>(
+                    vp + (long)src * ss + col);
+            } else {
+                kreg[i] = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+                vreg[i] = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+            }
+        }
+    };
+    auto write_k = [&](int buf) {
+#pragma unroll
+        for (int i = 0; i < KVECS; ++i) {
+            const int idx = tid + i * THREADS;
+            const int row = idx / (D / 8);
+            const int col = (idx % (D / 8)) * 8;
+            *reinterpret_cast<bf16x8*>(
+                &k_lds[buf * (KVBLK * KPAD) + row * KPAD + col]) = kreg[i];
+        }
+    };
+    auto write_v = [&](int buf) {
+#pragma unroll
+        for (int i = 0; i < KVECS; ++i) {
+            const int idx = tid + i * THREADS;
+            const int row = idx / (D / 8);
+            const int col = (idx % (D / 8)) * 8;
+            *reinterpret_cast<bf16x8*>(
+                &v_lds[buf * (KVBLK * VROW) + row * VROW +
+                       (col ^ ((row & 8) << 1))]) = vreg[i];
+        }
+    };
+
+    const int n_tiles = (Sk + KVBLK - 1) / KVBLK;
+    issue_loads(0);
+    write_k(0);
+    write_v(0);
+    if (n_tiles > 1) issue_loads(KVBLK);
+    __syncthreads();
+    for (int t = 0; t < n_tiles; ++t) {
+        const int p = t & 1;
+        const int kv0 = t * KVBLK;
+        // QK^T both frames: the K A-fragment is read from LDS ONCE per
+        // (kt, kk) and feeds both frames' MFMAs
+        f32x16 st[QT][2];
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int kt = 0; kt < 2; ++kt) {
+#pragma unroll
+            for (int qt = 0; qt < QT; ++qt)
+#pragma unroll
+                for (int r = 0; r < 16; ++r) st[qt][kt][r] = 0.f;
+#pragma unroll
+            for (int kk = 0; kk < KK; ++kk) {
+                bf16x8 afrag = *reinterpret_cast<const bf16x8*>(
+                    &k_lds[p * (KVBLK * KPAD) + (kt * 32 + l32) * KPAD +
+                           kk * 16 + hi * 8]);
+#pragma unroll
+                for (int qt = 0; qt < QT; ++qt)
+                    st[qt][kt] = mfma32x32x16(afrag, qfrag[qt][kk],
+                                              st[qt][kt]);
+            }
+        }
+        __builtin_amdgcn_s_setprio(0);
+        if (t + 1 < n_tiles) write_k(p ^ 1);
+
+        // softmax + P per frame
+        bf16x8 pfrag[QT][4];
+#pragma unroll
+        for (int qt = 0; qt < QT; ++qt) {
+            float mx = -3e30f;
+#pragma unroll
+            for (int kt = 0; kt < 2; ++kt)
+#pragma unroll
+                for (int r = 0; r < 16; ++r) {
+                    const int key =
+                        kv0 + kt * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+                    const float sv = (key < Sk) ? st[qt][kt][r] : -3e30f;
+                    st[qt][kt][r] = sv;
+                    mx = fmaxf(mx, sv);
+                }
+            mx = fmaxf(mx, __shfl_xor(mx, 32, 64));
+            const float mnew = fmaxf(m_run[qt], mx * scale2);
+            const float alpha = __builtin_amdgcn_exp2f(m_run[qt] - mnew);
+            m_run[qt] = mnew;
+            float ps = 0.f;
+#pragma unroll
+            for (int kt = 0; kt < 2; ++kt)
+#pragma unroll
+                for (int r = 0; r < 16; ++r) {
+                    const float pv_ = __builtin_amdgcn_exp2f(
+                        fmaf(st[qt][kt][r], scale2, -mnew));
+                    st[qt][kt][r] = pv_;
+                    ps += pv_;
+                }
+            ps += __shfl_xor(ps, 32, 64);
+            l_run[qt] = l_run[qt] * alpha + ps;
+            if (alpha != 1.f) {
+#pragma unroll
+                for (int n = 0; n < NV; ++n)
+#pragma unroll
+                    for (int r = 0; r < 16; ++r) o_acc[qt][n][r] *= alpha;
+            }
+#pragma unroll
+            for (int c = 0; c < 4; ++c) {
+                const f32x16& sv = st[qt][c >> 1];
+                const int rb = 8 * (c & 1);
+                unsigned int w0 = cvt_pk_bf16(sv[rb + 0], sv[rb + 1]);
+                unsigned int w1 = cvt_pk_bf16(sv[rb + 2], sv[rb + 3]);
+                unsigned int w2 = cvt_pk_bf16(sv[rb + 4], sv[rb + 5]);
+                unsigned int w3 = cvt_pk_bf16(sv[rb + 6], sv[rb + 7]);
+                auto r02 =
+                    __builtin_amdgcn_permlane32_swap(w0, w2, false, false);
+                auto r13 =
+                    __builtin_amdgcn_permlane32_swap(w1, w3, false, false);
+                unsigned int dd[4] = {(unsigned int)r02[0],
+                                      (unsigned int)r13[0],
+                                      (unsigned int)r02[1],
+                                      (unsigned int)r13[1]};
+                pfrag[qt][c] = *reinterpret_cast<bf16x8*>(dd);
+            }
+        }
+        if (t + 1 < n_tiles) write_v(p ^ 1);
+        if (t + 2 < n_tiles) issue_loads((t + 2) * KVBLK);
+
+        // PV both frames: each V tr-read pair feeds both frames' MFMAs
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int c = 0; c < 4; ++c) {
+#pragma unroll
+            for (int n = 0; n < NV; ++n) {
+                s16x4 alo = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+                    (LDS_P s16x4*)&v_lds[p * (KVBLK * VROW) + trb +
+                                         c * (16 * VROW) + n * 32]);
+                s16x4 ahi = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+                    (LDS_P s16x4*)&v_lds[p * (KVBLK * VROW) + trb +
+                                         c * (16 * VROW) + 4 * VROW +
+                                         n * 32]);
+                bf16x8 va =
+                    __builtin_shufflevector(alo, ahi, 0, 1, 2, 3, 4, 5, 6, 7);
+#pragma unroll
+                for (int qt = 0; qt < QT; ++qt)
+                    o_acc[qt][n] = mfma32x32x16(va, pfrag[qt][c],
+                                                o_acc[qt][n]);
+            }
+        }
+        __builtin_amdgcn_s_setprio(0);
+        __syncthreads();
+    }
+
+#pragma unroll
+    for (int qt = 0; qt < QT; ++qt) {
+        const int row = q0 + qt * 32 + l32;
+        if (row < S) {
+            const float inv_l = (l_run[qt] > 0.f) ? 1.f / l_run[qt] : 0.f;
+#pragma unroll
+            for (int n = 0; n < NV; ++n)
+#pragma unroll
+                for (int r2 = 0; r2 < 4; ++r2) {
+                    const int dim0 = n * 32 + 8 * r2 + 4 * hi;
+                    unsigned short pack[4];
+#pragma unroll
+                    for (int j = 0; j < 4; ++j)
+                        pack[j] = __bfloat16_as_ushort(
+                            f2bf(o_acc[qt][n][r2 * 4 + j] * inv_l));
+                    *reinterpret_cast<unsigned long long*>(
+                        op + (long)row * ss + dim0) =
+                        *reinterpret_cast<unsigned long long*>(pack);
+                }
+        }
+    }
+}
+
 #define HIP_CHECK(x) do { hipError_t e = (x); if (e) { \
     printf("HIP error %d at %d\n", e, __LINE__); exit(1); } } while (0)
 
@@ -491,13 +744,25 @@ static void launch(int var, const bf16* q, const bf16* k, const bf16* v,
         hipLaunchKernelGGL((attn_v5_kernel<D, 4>), grid, blk, 0, 0, q, k, v,
                            o, S, S, scale, H, inv_s, s2);
         break;
-    default:
+    case 5:
         hipLaunchKernelGGL((attn_v5_kernel<D, 5>), grid, blk, 0, 0, q, k, v,
                            o, S, S, scale, H, inv_s, s2);
+        break;
+    default:
+        // var6: two-q-tile wave — a D=64 remedy; at D=128 time v5a so the
+        // interleaved rounds stay aligned
+        if (D == 64) {
+            dim3 g2(((S + 511) / 512) * B * H);
+            hipLaunchKernelGGL((attn_d64x2_kernel<64>), g2, blk, 0, 0, q, k,
+                               v, o, S, S, scale, H);
+        } else {
+            hipLaunchKernelGGL((attn_v5_kernel<D, 1>), grid, blk, 0, 0, q, k,
+                               v, o, S, S, scale, H, inv_s, s2);
+        }
     }
 }
 
-constexpr int NVAR = 6;
+constexpr int NVAR = 7;
 
 template <int D>
 static int check_correct(int B, int H, int S, bool spike) {
