@@ -85,6 +85,14 @@ def main(argv=None):
     if args.model in ("sd15", "sdxl") and on_gpu:
         # MIOpen conv algo search (cached in-process) for the UNet families
         torch.backends.cudnn.benchmark = True
+    if on_gpu:
+        # pre-tuned hipBLASLt algorithm selection (measured -1.0% flagship;
+        # PA_NO_TUNABLEOP=1 disables)
+        from comfyui_parallelanything_amd.utils.tunable import (
+            enable_tuned_gemms,
+        )
+
+        enable_tuned_gemms()
     make, make_inputs = MODELS[args.model]
     model = make(dev=dev, dtype=dtype, tiny=tiny)
     # replicate(): rank0's weights to every replica, flat bucketed RCCL bcast

@@ -64,6 +64,10 @@ def main(argv=None):
     if tiny and dtype != torch.float32:
         dtype = torch.float32
 
+    if torch.cuda.is_available():
+        from .utils.tunable import enable_tuned_gemms
+
+        enable_tuned_gemms()
     chain = DeviceChain.from_list(
         [make_entry(d, p) for d, p in zip(devices, pcts)]
     )
