@@ -37,6 +37,7 @@ def create_app(
     tiny: bool = False,
     microbatches: int = 1,
     hip_graphs: bool = False,
+    fp8: bool = False,
 ):
     from fastapi import FastAPI, HTTPException
     from pydantic import BaseModel
@@ -46,11 +47,19 @@ def create_app(
     if tiny or not torch.cuda.is_available():
         tiny, dtype = True, torch.float32
 
+    if torch.cuda.is_available():
+        from .utils.tunable import enable_tuned_gemms
+
+        enable_tuned_gemms()
     chain = DeviceChain.from_list(
         [make_entry(d, p) for d, p in zip(devices, percents)]
     )
     make, make_inputs = MODELS[model_name]
     model = make(dev=chain.lead, dtype=dtype, tiny=tiny)
+    if fp8:
+        from .models.quant import quantize_fp8
+
+        quantize_fp8(model)
     engine = ParallelEngine(chain, use_hip_graphs=hip_graphs)
     engine.setup(model)
     configure_pipeline(engine, microbatches=microbatches)
@@ -79,6 +88,7 @@ def create_app(
             "devices": list(chain.devices),
             "weights": list(chain.weights),
             "dtype": str(dtype),
+            "fp8": fp8,
         }
 
     @app.post("/generate")
@@ -133,13 +143,15 @@ def main(argv=None) -> None:
     ap.add_argument("--tiny", action="store_true")
     ap.add_argument("--microbatches", type=int, default=1)
     ap.add_argument("--hip-graphs", action="store_true")
+    ap.add_argument("--fp8", action="store_true",
+                    help="e4m3fn serving mode (quantize_fp8; GPU only)")
     args = ap.parse_args(argv)
     devices = args.devices.split(",") if args.devices else None
     percents = ([float(p) for p in args.percent.split(",")]
                 if args.percent else None)
     app = create_app(args.model, devices, percents, tiny=args.tiny,
                      microbatches=args.microbatches,
-                     hip_graphs=args.hip_graphs)
+                     hip_graphs=args.hip_graphs, fp8=args.fp8)
     uvicorn.run(app, host=args.host, port=args.port)
 
 
