@@ -234,3 +234,36 @@ def test_fp8_model_uses_fused_ln(monkeypatch):
         out = m(x, t, context=c, **kw)
     assert torch.isfinite(out.float()).all()
     assert calls["n"] > 0, "fused LN+quant path not taken"
+
+
+def test_fnuz_to_ocp_reencode_on_gpu():
+    """fnuz -> OCP re-encode on DEVICE tensors (round-1 gap: CPU-only
+    coverage). Values must survive the format change (different exponent
+    bias makes a bit-cast wrong)."""
+    from comfyui_parallelanything_amd.parallel.fp8 import (
+        sanitize_param_dtype, to_ocp_fp8,
+    )
+
+    if not hasattr(torch, "float8_e4m3fnuz"):
+        pytest.skip("no fnuz dtype in this torch")
+    torch.manual_seed(7)
+    vals = torch.randn(256, device="cuda").clamp(-8, 8)
+    fnuz = vals.to(torch.float8_e4m3fnuz)
+    ocp = to_ocp_fp8(fnuz)
+    assert ocp.dtype == torch.float8_e4m3fn and ocp.is_cuda
+    # both encodings decode to (approximately) the same reals
+    torch.testing.assert_close(
+        ocp.float(), fnuz.float(), rtol=0.07, atol=0.02
+    )
+    # sanitize keeps fp8 on GPU (gfx950 native) and normalizes the format
+    s = sanitize_param_dtype(fnuz, "cuda:0")
+    assert s.dtype == torch.float8_e4m3fn
+    # replication of a module carrying an fnuz buffer lands OCP on device
+    from comfyui_parallelanything_amd.parallel.replicate import (
+        replicate_module,
+    )
+
+    m = torch.nn.Linear(8, 8).cuda().to(torch.bfloat16)
+    m.register_buffer("w8", fnuz.clone())
+    rep = replicate_module(m, "cuda:0", force_copy=True)
+    assert rep.w8.dtype == torch.float8_e4m3fn and rep.w8.is_cuda
