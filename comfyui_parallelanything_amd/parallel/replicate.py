@@ -23,7 +23,7 @@ from __future__ import annotations
 
 import copy
 import logging
-from typing import Iterable, List, Optional
+from typing import List, Optional
 
 import torch
 import torch.distributed as dist
