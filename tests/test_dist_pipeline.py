@@ -82,3 +82,33 @@ def _flux_worker3(rank, world, tmpdir):
 
 def test_dist_pipeline_world3(tmp_path):
     mp.spawn(_flux_worker3, args=(3, str(tmp_path)), nprocs=3, join=True)
+
+
+def _wan_i2v_worker4(rank, world, tmpdir):
+    from comfyui_parallelanything_amd.models.registry import (
+        make_wan_i2v, wan_i2v_inputs,
+    )
+    from comfyui_parallelanything_amd.parallel.dist_pipeline import (
+        install_dist_pipeline, uninstall_dist_pipeline,
+    )
+
+    info = _init(rank, world, tmpdir)
+    torch.manual_seed(0)
+    model = make_wan_i2v(tiny=True, dtype=torch.float32)
+    x, t, c, kw = wan_i2v_inputs(1, tiny=True, dtype=torch.float32)
+    ref = model(x, t, context=c, **kw)
+    n = install_dist_pipeline(model, info)
+    assert n > 0  # transformer_blocks sharded
+    out = model(x, t, context=c, **kw)
+    if rank == 0:
+        torch.testing.assert_close(out, ref, rtol=1e-5, atol=1e-6)
+    assert uninstall_dist_pipeline(model) == n
+    out2 = model(x, t, context=c, **kw)
+    torch.testing.assert_close(out2, ref, rtol=0, atol=0)
+    dist.destroy_process_group()
+
+
+def test_dist_pipeline_world4_wan_i2v(tmp_path):
+    """4-rank layer sharding of the I2V video DiT (image_cond kwarg flows
+    locally on every rank; hidden state hands off by p2p)."""
+    mp.spawn(_wan_i2v_worker4, args=(4, str(tmp_path)), nprocs=4, join=True)
