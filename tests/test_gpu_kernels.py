@@ -349,3 +349,39 @@ def test_timestep_embed_mlp_matches_composed_path():
     fused = emb.forward_timestep(t)
     composed = emb(ops.timestep_embedding(t, 256).to(torch.bfloat16))
     _cmp(fused, composed.float(), 3e-2, 3e-2, "ts mlp fused-vs-composed")
+
+
+@pytest.mark.parametrize("name", ["flux", "sd15", "zimage", "sd3", "wan_i2v"])
+def test_whole_model_gpu_vs_cpu_reference(name):
+    """End-to-end composition check: the bf16 HIP-kernel path on GPU vs
+    the SAME tiny model in fp32 on CPU (reference ops). Per-op numerics
+    tests bound each kernel; this bounds their COMPOSITION through a full
+    forward (block stacks, fused qkv/norm/rope paths, MIOpen convs)."""
+    from comfyui_parallelanything_amd.models.registry import MODELS
+
+    make, inputs = MODELS[name]
+    torch.manual_seed(0)
+    m_ref = make(dev="cpu", dtype=torch.float32, tiny=True)
+    torch.manual_seed(0)
+    m_gpu = make(dev="cuda", dtype=torch.bfloat16, tiny=True)
+    # same random init (same seed) — verify before trusting the comparison
+    p_ref = next(m_ref.parameters())
+    p_gpu = next(m_gpu.parameters())
+    torch.testing.assert_close(
+        p_gpu.float().cpu(), p_ref.float(), rtol=5e-3, atol=5e-3
+    )
+    x, t, c, kw = inputs(2, tiny=True, dtype=torch.float32)
+    with torch.no_grad():
+        ref = m_ref(x, t, context=c, **kw).float()
+        out = m_gpu(
+            x.cuda().bfloat16(), t.cuda(),
+            context=c.cuda().bfloat16(),
+            **{k: (v.cuda().bfloat16() if isinstance(v, torch.Tensor) else v)
+               for k, v in kw.items()},
+        ).float().cpu()
+    # bf16 drift compounds across blocks: bound the relative l2 error and
+    # require strong agreement in direction
+    rel = (out - ref).norm() / ref.norm()
+    assert rel < 0.15, f"{name}: whole-model rel-l2 {rel:.4f}"
+    corr = torch.corrcoef(torch.stack([out.flatten(), ref.flatten()]))[0, 1]
+    assert corr > 0.99, f"{name}: correlation {corr:.4f}"
