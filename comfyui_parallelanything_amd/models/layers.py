@@ -76,10 +76,37 @@ class FusedMLP(nn.Module):
         self.up = GELULinear(dim_in, dim_mid)
         self.down = nn.Linear(dim_mid, dim_out, bias=True)
 
+    def _down_fp8(self, y):
+        """fp8 mode: GELU + e4m3 cast in ONE kernel feeding the down GEMM
+        (no standalone quant pass); None when the fused path is off."""
+        if (hasattr(self.down, "gelu_quant") and y.is_cuda
+                and y.dtype == torch.bfloat16
+                and ops.hip_available("gelu_fp8")):
+            x8, s_used = self.down.gelu_quant(y)
+            return self.down.mm_fp8(x8, s_used)
+        return None
+
     def forward(self, x):
+        up_lin = getattr(self.up, "lin", None)
+        if hasattr(up_lin, "mm_fp8"):
+            y = up_lin(x)  # fp8 GEMM (bf16 out), GELU NOT yet applied
+            out = self._down_fp8(y)
+            if out is not None:
+                return out
+            return self.down(ops.gelu_tanh(y) if y.is_cuda else
+                             torch.nn.functional.gelu(y, approximate="tanh"))
         return self.down(self.up(x))
 
     def forward_ln(self, x, scale, shift):
+        up_lin = getattr(self.up, "lin", None)
+        ln_fwd = getattr(up_lin, "forward_ln", None)
+        if ln_fwd is not None:
+            y = ln_fwd(x, scale, shift)  # fused LN+quant -> fp8 GEMM
+            out = self._down_fp8(y)
+            if out is not None:
+                return out
+            return self.down(ops.gelu_tanh(y) if y.is_cuda else
+                             torch.nn.functional.gelu(y, approximate="tanh"))
         return self.down(self.up.forward_ln(x, scale, shift))
 
 
@@ -315,6 +342,7 @@ class SingleStreamBlock(nn.Module):
             mod, _ = self.modulation(vec)
         qkv_lin = self.linear1_qkv
         mlp_lin = getattr(self.linear1_mlp, "lin", None)
+        mlp_fp8 = None  # (x8, scale) for the fused gelu+quant hand-off
         if (hasattr(qkv_lin, "ln_quant") and hasattr(mlp_lin, "mm_fp8")
                 and x.is_cuda and x.dtype == torch.bfloat16
                 and ops.hip_available("layer_norm_mod_fp8")):
@@ -323,7 +351,14 @@ class SingleStreamBlock(nn.Module):
             # one LN + two standalone quant passes.
             x8, s_used = qkv_lin.ln_quant(x, mod.scale, mod.shift)
             qkv = qkv_lin.mm_fp8(x8, s_used)
-            mlp_act = ops.gelu_tanh(mlp_lin.mm_fp8(x8, s_used))
+            y_up = mlp_lin.mm_fp8(x8, s_used)
+            if (hasattr(self.linear2_mlp, "gelu_quant")
+                    and ops.hip_available("gelu_fp8")):
+                # GELU + e4m3 cast in one kernel feeding linear2_mlp
+                mlp_fp8 = self.linear2_mlp.gelu_quant(y_up)
+                mlp_act = None
+            else:
+                mlp_act = ops.gelu_tanh(y_up)
         else:
             x_in = ops.layer_norm_mod(x, mod.scale, mod.shift)
             qkv = self.linear1_qkv(x_in)
@@ -334,7 +369,9 @@ class SingleStreamBlock(nn.Module):
             q, k, self.norm.query_norm.scale, self.norm.key_norm.scale, pe
         )
         attn = ops.attention_bshd(q, k, v, self.attn.scale).flatten(2)
-        if x.is_cuda and isinstance(self.linear2_mlp, nn.Linear):
+        if mlp_fp8 is not None:
+            out = self.linear2_attn(attn) + self.linear2_mlp.mm_fp8(*mlp_fp8)
+        elif x.is_cuda and isinstance(self.linear2_mlp, nn.Linear):
             # second GEMM accumulates into the first's output (beta=1
             # epilogue) — no separate elementwise add
             acc = self.linear2_attn(attn).reshape(-1, hidden)

@@ -124,6 +124,17 @@ class FP8Linear(nn.Module):
                                     self.a_amax, self.x_scale_used)
         return x8, self.x_scale_used
 
+    def gelu_quant(self, y: torch.Tensor):
+        """tanh-GELU with the fp8 cast fused (this Linear consumes the
+        GELU output): returns (x8, scale_used) for mm_fp8."""
+        from .. import ops
+
+        if not self._warm:
+            g = torch.nn.functional.gelu(y.float(), approximate="tanh")
+            self._warm_from(g.abs().amax().reshape(1).float())
+        x8 = ops.gelu_fp8(y, self.x_scale, self.a_amax, self.x_scale_used)
+        return x8, self.x_scale_used
+
     def forward_ln(self, x: torch.Tensor, mscale: torch.Tensor,
                    mshift: torch.Tensor) -> torch.Tensor:
         """layer_norm_mod -> this Linear, with the quant fused into the LN

@@ -291,6 +291,14 @@ def quant_fp8(x: torch.Tensor, scale: torch.Tensor,
                          scale if scale_used is None else scale_used)
 
 
+def gelu_fp8(x, scale, amax_buf, scale_used) -> torch.Tensor:
+    """tanh-GELU emitting e4m3fn with quant_fp8's delayed-scaling contract
+    (fp8 serving mode: the MLP-down GEMM consumes the output directly,
+    no standalone quant pass). GPU-only."""
+    ext = _require_ext("gelu_fp8")
+    return ext.gelu_fp8(x, scale, amax_buf, scale_used)
+
+
 def gelu_tanh(x: torch.Tensor) -> torch.Tensor:
     if x.is_cuda:
         ext = _require_ext("gelu_tanh")

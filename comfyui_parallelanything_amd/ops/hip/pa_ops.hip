@@ -387,6 +387,68 @@ __global__ void gate_residual_bf16_kernel(const bf16* __restrict__ res,
 
 // Vectorized tanh-GELU (bf16 short8): the MLP activation between the two
 // hipBLASLt GEMMs ([B, S, 4*hidden] tensors; memory-bound).
+// tanh-GELU with the bf16 -> e4m3fn cast fused (fp8 serving mode): the
+// MLP-down projection consumes the GELU output, so emitting fp8 directly
+// removes its standalone quant pass (same delayed-scaling epilogue as
+// quant_fp8: one amax atomic + counter hit per block, last block writes
+// the next scale and snapshots the used one).
+__global__ void gelu_fp8_kernel(const bf16* __restrict__ x,
+                                unsigned char* __restrict__ out, long total8,
+                                float* __restrict__ scale,
+                                float* __restrict__ amax_buf,
+                                float* scale_used) {
+    const long stride = (long)gridDim.x * blockDim.x;
+    const short8* xv = reinterpret_cast<const short8*>(x);
+    const float s_entry = scale[0];
+    const float inv_s = 1.0f / s_entry;
+    float local_amax = 0.f;
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total8;
+         i += stride) {
+        short8 v = xv[i];
+        unsigned char pack[8];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            const float f = bf2f(__ushort_as_bfloat16((unsigned short)v[j]));
+            const float c = 0.7978845608028654f * (f + 0.044715f * f * f * f);
+            const float g = 0.5f * f * (1.f + tanhf(c));
+            local_amax = fmaxf(local_amax, fabsf(g));
+            const float qv = fminf(fmaxf(g * inv_s, -448.f), 448.f);
+            pack[j] = (unsigned char)__hip_cvt_float_to_fp8(
+                qv, __HIP_SATFINITE, __HIP_E4M3);
+        }
+        *reinterpret_cast<unsigned long long*>(&out[i * 8]) =
+            *reinterpret_cast<unsigned long long*>(pack);
+    }
+    __shared__ float scratch[8];
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1)
+        local_amax = fmaxf(local_amax, __shfl_xor(local_amax, off, 64));
+    const int lane = threadIdx.x & 63;
+    const int wid = threadIdx.x >> 6;
+    if (lane == 0) scratch[wid] = local_amax;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        float m = 0.f;
+        for (int i = 0; i < (int)(blockDim.x / 64); ++i)
+            m = fmaxf(m, scratch[i]);
+        atomicMax(reinterpret_cast<unsigned int*>(amax_buf),
+                  __float_as_uint(m));
+        __threadfence();
+        const unsigned int done = atomicAdd(
+            reinterpret_cast<unsigned int*>(&amax_buf[1]), 1u);
+        if (done == gridDim.x - 1) {
+            __threadfence();
+            reinterpret_cast<unsigned int*>(amax_buf)[1] = 0u;
+            const float next = amax_buf[0] * 0.999f;
+            amax_buf[0] = next;
+            if (scale_used != scale) scale_used[0] = s_entry;
+            scale[0] = fmaxf(next / 448.f, 1e-12f);
+            __threadfence();
+        }
+    }
+}
+
+
 __global__ void gelu_tanh_bf16_kernel(const bf16* __restrict__ x,
                                       bf16* __restrict__ out, long total8,
                                       long rows, long w8, long in_stride8) {
@@ -1683,6 +1745,24 @@ __global__ __launch_bounds__(512, 2) void attn_fwd_v4_kernel(
 }
 
 
+at::Tensor gelu_fp8(at::Tensor x, at::Tensor scale, at::Tensor amax_buf,
+                    at::Tensor scale_used) {
+    CHECK_GPU(x);
+    TORCH_CHECK(x.scalar_type() == at::kBFloat16, "gelu_fp8: bf16 input");
+    auto xc = x.contiguous();
+    TORCH_CHECK((xc.numel() % 8) == 0, "gelu_fp8: numel % 8 == 0");
+    TORCH_CHECK(amax_buf.numel() >= 2, "gelu_fp8: amax_buf needs counter");
+    auto out = at::empty(xc.sizes(), xc.options().dtype(at::kFloat8_e4m3fn));
+    const long total8 = xc.numel() / 8;
+    const int blocks = (int)std::min<long>((total8 + 255) / 256, 4096);
+    hipLaunchKernelGGL(gelu_fp8_kernel, dim3(blocks), dim3(256), 0,
+                       cur_stream(), (const bf16*)xc.data_ptr(),
+                       (unsigned char*)out.data_ptr(), total8,
+                       scale.data_ptr<float>(), amax_buf.data_ptr<float>(),
+                       scale_used.data_ptr<float>());
+    return out;
+}
+
 at::Tensor gelu_tanh(at::Tensor x) {
     CHECK_GPU(x);
     // accept a 2-D-decomposable strided view: last dim contiguous, all
@@ -1926,6 +2006,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("attn_fwd_bshd_split", &attn_fwd_bshd_split,
           "Attention with per-stream split outputs [B,:split]/[B,split:]");
     m.def("gelu_tanh", &gelu_tanh, "Vectorized tanh-GELU (gfx950)");
+    m.def("gelu_fp8", &gelu_fp8,
+          "tanh-GELU with fused e4m3fn quant + delayed scaling (gfx950)");
     m.def("quant_fp8", &quant_fp8,
           "Fused bf16->e4m3fn quant with running amax (gfx950)");
     m.def("timestep_embed_mlp", &timestep_embed_mlp,

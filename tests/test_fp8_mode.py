@@ -267,3 +267,56 @@ def test_fnuz_to_ocp_reencode_on_gpu():
     m.register_buffer("w8", fnuz.clone())
     rep = replicate_module(m, "cuda:0", force_copy=True)
     assert rep.w8.dtype == torch.float8_e4m3fn and rep.w8.is_cuda
+
+
+def test_gelu_fp8_fused():
+    """Fused tanh-GELU + e4m3 quant == gelu then quantize, same
+    delayed-scaling contract."""
+    from comfyui_parallelanything_amd import ops
+
+    if not ops.hip_available("gelu_fp8"):
+        pytest.skip("no gelu_fp8 in extension")
+    torch.manual_seed(8)
+    y = torch.randn(64, 2048, device="cuda", dtype=torch.bfloat16) * 2
+    g = torch.nn.functional.gelu(y.float(), approximate="tanh")
+    true_amax = g.abs().amax().item()
+    scale = torch.tensor([true_amax / 448.0], device="cuda")
+    amax = torch.zeros(2, device="cuda")
+    used = torch.zeros(1, device="cuda")
+    x8 = ops.gelu_fp8(y, scale, amax, used)
+    deq = x8.float() * used
+    rel = (deq - g).norm() / g.norm()
+    assert rel < 0.05, f"fused gelu+quant error {rel:.4f}"
+    assert abs(amax[0].item() - true_amax * 0.999) < 2e-2
+    assert amax[1].item() == 0.0
+
+
+def test_fp8_fusedmlp_uses_gelu_fp8(monkeypatch):
+    """Quantized FusedMLP routes GELU through the fused gelu+quant path
+    and matches the composed fp8 computation."""
+    from comfyui_parallelanything_amd import ops
+    from comfyui_parallelanything_amd.models.layers import FusedMLP
+    from comfyui_parallelanything_amd.models.quant import (
+        FP8Linear, _supports_scaled_mm, quantize_fp8,
+    )
+
+    if not _supports_scaled_mm() or not ops.hip_available("gelu_fp8"):
+        pytest.skip("no fp8 gelu path")
+    torch.manual_seed(9)
+    m = FusedMLP(256, 1024, 256).cuda().to(torch.bfloat16)
+    x = torch.randn(2, 32, 256, device="cuda", dtype=torch.bfloat16)
+    ref = m(x).float()  # bf16 path before quantization
+    quantize_fp8(m, min_features=256)
+    assert isinstance(m.down, FP8Linear)
+    calls = {"n": 0}
+    orig = FP8Linear.gelu_quant
+
+    def spy(self, *a, **k):
+        calls["n"] += 1
+        return orig(self, *a, **k)
+
+    monkeypatch.setattr(FP8Linear, "gelu_quant", spy)
+    out = m(x).float()
+    assert calls["n"] > 0, "fused gelu+quant path not taken"
+    rel = (out - ref).norm() / ref.norm()
+    assert rel < 0.08, f"fp8 FusedMLP error {rel:.4f}"
