@@ -357,19 +357,15 @@ def test_whole_model_gpu_vs_cpu_reference(name):
     the SAME tiny model in fp32 on CPU (reference ops). Per-op numerics
     tests bound each kernel; this bounds their COMPOSITION through a full
     forward (block stacks, fused qkv/norm/rope paths, MIOpen convs)."""
+    import copy
+
     from comfyui_parallelanything_amd.models.registry import MODELS
 
     make, inputs = MODELS[name]
-    torch.manual_seed(0)
     m_ref = make(dev="cpu", dtype=torch.float32, tiny=True)
-    torch.manual_seed(0)
-    m_gpu = make(dev="cuda", dtype=torch.bfloat16, tiny=True)
-    # same random init (same seed) — verify before trusting the comparison
-    p_ref = next(m_ref.parameters())
-    p_gpu = next(m_gpu.parameters())
-    torch.testing.assert_close(
-        p_gpu.float().cpu(), p_ref.float(), rtol=5e-3, atol=5e-3
-    )
+    # identical weights BY CONSTRUCTION (CPU and CUDA RNGs differ per
+    # seed, so re-making on cuda would produce a different init)
+    m_gpu = copy.deepcopy(m_ref).to("cuda", torch.bfloat16)
     x, t, c, kw = inputs(2, tiny=True, dtype=torch.float32)
     with torch.no_grad():
         ref = m_ref(x, t, context=c, **kw).float()
