@@ -410,7 +410,15 @@ __global__ void gelu_fp8_kernel(const bf16* __restrict__ x,
         for (int j = 0; j < 8; ++j) {
             const float f = bf2f(__ushort_as_bfloat16((unsigned short)v[j]));
             const float c = 0.7978845608028654f * (f + 0.044715f * f * f * f);
-            const float g = 0.5f * f * (1.f + tanhf(c));
+            // tanh via the HARDWARE exp2 (v_exp_f32): libm tanhf measured
+            // only 2.7 TB/s on this pass (rocprof r02 fp8 mix). Negative-
+            // exponent form so exp2 never overflows (E=inf would NaN):
+            // g = 0.5*f*(1+tanh(c)); with En = exp2(-2|c|log2e) in (0,1],
+            // tanh(|c|) = (1-En)/(1+En) -> g = f * r or f * (1-r).
+            const float En =
+                __builtin_amdgcn_exp2f(-2.8853900817779268f * fabsf(c));
+            const float r = 1.f / (1.f + En);
+            const float g = f * (c >= 0.f ? r : 1.f - r);
             local_amax = fmaxf(local_amax, fabsf(g));
             const float qv = fminf(fmaxf(g * inv_s, -448.f), 448.f);
             pack[j] = (unsigned char)__hip_cvt_float_to_fp8(
@@ -466,7 +474,15 @@ __global__ void gelu_tanh_bf16_kernel(const bf16* __restrict__ x,
         for (int j = 0; j < 8; ++j) {
             const float f = bf2f(__ushort_as_bfloat16((unsigned short)v[j]));
             const float c = 0.7978845608028654f * (f + 0.044715f * f * f * f);
-            const float g = 0.5f * f * (1.f + tanhf(c));
+            // tanh via the HARDWARE exp2 (v_exp_f32): libm tanhf measured
+            // only 2.7 TB/s on this pass (rocprof r02 fp8 mix). Negative-
+            // exponent form so exp2 never overflows (E=inf would NaN):
+            // g = 0.5*f*(1+tanh(c)); with En = exp2(-2|c|log2e) in (0,1],
+            // tanh(|c|) = (1-En)/(1+En) -> g = f * r or f * (1-r).
+            const float En =
+                __builtin_amdgcn_exp2f(-2.8853900817779268f * fabsf(c));
+            const float r = 1.f / (1.f + En);
+            const float g = f * (c >= 0.f ? r : 1.f - r);
             o[j] = (short)__bfloat16_as_ushort(f2bf(g));
         }
         ov[i] = o;
