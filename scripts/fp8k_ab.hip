@@ -217,7 +217,7 @@ int main(int argc, char** argv) {
     unsigned char* o;
     float *scale, *amax, *scratch;
     HIP_CHECK(hipMalloc(&x, n * 2));
-    HIP_CHECK(hipMalloc(&o, n));
+    HIP_CHECK(hipMalloc(&o, n * 2));  // v7 stores bf16 (2 bytes/elem)
     HIP_CHECK(hipMalloc(&scale, 4));
     HIP_CHECK(hipMalloc(&amax, 8));
     HIP_CHECK(hipMalloc(&scratch, 4 * 65536));
@@ -238,7 +238,8 @@ int main(int argc, char** argv) {
         HIP_CHECK(hipMemcpy(scale, &one, 4, hipMemcpyHostToDevice));
         HIP_CHECK(hipMemset(amax, 0, 8));
     }
-    const double gb = (n * 2 + n) / 1e9;
+    const double gb = (n * 2 + n) / 1e9;      // 16B read + 8B write
+    const double gb16 = (n * 2 + n * 2) / 1e9;  // v7: 16B read + 16B write
     auto run = [&](int var, int grid) {
         dim3 g(grid), b(256);
         switch (var) {
@@ -290,6 +291,6 @@ int main(int argc, char** argv) {
         }
     for (int i = 0; i < NV; ++i)
         printf("%-26s %8.1f us  %6.2f TB/s\n", vs[i].name, best[i] * 1e3,
-               gb / best[i]);
+               (vs[i].var == 7 ? gb16 : gb) / best[i]);
     return 0;
 }
