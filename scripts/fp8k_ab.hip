@@ -28,6 +28,76 @@ typedef __attribute__((ext_vector_type(8))) short short8;
 
 __device__ __forceinline__ float bf2f(bf16 v) { return __bfloat162float(v); }
 
+// v8: 4-deep ILP + CONDITIONAL atomicMax (skip when not a new max) +
+// the highest-blockIdx block writes the next scale from the running amax
+// (no counter chain; delayed scaling tolerates a fraction-of-a-call lag)
+__global__ void gelu_fp8_v8(const bf16* __restrict__ x,
+                            unsigned char* __restrict__ out, long total8,
+                            float* __restrict__ scale,
+                            float* __restrict__ amax_buf,
+                            float* __restrict__ scale_used) {
+    const long stride = (long)gridDim.x * blockDim.x;
+    const short8* xv = reinterpret_cast<const short8*>(x);
+    const float s_entry = scale[0];
+    const float inv_s = 1.0f / s_entry;
+    float local_amax = 0.f;
+    const long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    for (long i = i0; i < total8; i += stride * 4) {
+        short8 v[4];
+        long idx[4];
+#pragma unroll
+        for (int h = 0; h < 4; ++h) {
+            idx[h] = i + h * stride;
+            if (idx[h] < total8) v[h] = xv[idx[h]];
+        }
+#pragma unroll
+        for (int h = 0; h < 4; ++h) {
+            if (idx[h] >= total8) break;
+            unsigned char pack[8];
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                const float f =
+                    bf2f(__ushort_as_bfloat16((unsigned short)v[h][j]));
+                const float c =
+                    0.7978845608028654f * (f + 0.044715f * f * f * f);
+                const float En = __builtin_amdgcn_exp2f(
+                    -2.8853900817779268f * fabsf(c));
+                const float r = 1.f / (1.f + En);
+                const float g = f * (c >= 0.f ? r : 1.f - r);
+                local_amax = fmaxf(local_amax, fabsf(g));
+                const float qv = fminf(fmaxf(g * inv_s, -448.f), 448.f);
+                pack[j] = (unsigned char)__hip_cvt_float_to_fp8(
+                    qv, __HIP_SATFINITE, __HIP_E4M3);
+            }
+            *reinterpret_cast<unsigned long long*>(&out[idx[h] * 8]) =
+                *reinterpret_cast<unsigned long long*>(pack);
+        }
+    }
+    __shared__ float red[8];
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1)
+        local_amax = fmaxf(local_amax, __shfl_xor(local_amax, off, 64));
+    const int lane = threadIdx.x & 63;
+    const int wid = threadIdx.x >> 6;
+    if (lane == 0) red[wid] = local_amax;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        float m = 0.f;
+        for (int i = 0; i < (int)(blockDim.x / 64); ++i)
+            m = fmaxf(m, red[i]);
+        if (m > amax_buf[0])  // plain read; atomic only on a new max
+            atomicMax(reinterpret_cast<unsigned int*>(amax_buf),
+                      __float_as_uint(m));
+        if (blockIdx.x == gridDim.x - 1) {
+            __threadfence();
+            const float next = fmaxf(amax_buf[0], m) * 0.999f;
+            amax_buf[0] = next;
+            scale_used[0] = s_entry;
+            scale[0] = fmaxf(next / 448.f, 1e-12f);
+        }
+    }
+}
+
 template <int VAR>
 __global__ void gelu_fp8_ab(const bf16* __restrict__ x,
                             unsigned char* __restrict__ out, long total8,
@@ -85,7 +155,8 @@ __global__ void gelu_fp8_ab(const bf16* __restrict__ x,
                     *reinterpret_cast<unsigned long long*>(pack);
             }
         }
-        if (local_amax < 0.f) out[0] = 1;
+        scratch_amax[(blockIdx.x * blockDim.x + threadIdx.x) & 65535] =
+            local_amax;
         return;
     } else if (VAR == 7) {
         // gelu compute but STORE BF16 (16B) — isolates the 8B-store cost
@@ -106,7 +177,8 @@ __global__ void gelu_fp8_ab(const bf16* __restrict__ x,
             }
             *reinterpret_cast<short8*>(&out[i * 16]) = ovv;
         }
-        if (local_amax < 0.f) out[0] = 1;
+        scratch_amax[(blockIdx.x * blockDim.x + threadIdx.x) & 65535] =
+            local_amax;
         return;
     }
     if (VAR == 4) {
@@ -161,8 +233,10 @@ __global__ void gelu_fp8_ab(const bf16* __restrict__ x,
         }
     }
     if (VAR == 2 || VAR >= 5) {
-        // no epilogue: keep local_amax alive without publishing
-        if (local_amax < 0.f) out[0] = 1;
+        // no ATOMIC epilogue — but the amax chain must stay live (a sign
+        // test lets the compiler DCE the whole fmax chain): plain store
+        scratch_amax[(blockIdx.x * blockDim.x + threadIdx.x) & 65535] =
+            local_amax;
         return;
     }
     __shared__ float red[8];
@@ -257,17 +331,18 @@ int main(int argc, char** argv) {
                                    scale, amax, scratch); break;
         case 7: hipLaunchKernelGGL(gelu_fp8_ab<7>, g, b, 0, 0, x, o, total8,
                                    scale, amax, scratch); break;
+        case 8: hipLaunchKernelGGL(gelu_fp8_v8, g, b, 0, 0, x, o, total8,
+                                   scale, amax, scratch); break;
         default: hipLaunchKernelGGL(gelu_fp8_ab<4>, g, b, 0, 0, x, o, total8,
                                     scale, amax, scratch); break;
         }
     };
     struct V { int var; int grid; const char* name; };
-    V vs[] = {{0, 4096, "v0 grid4096+atomics"},
-              {2, 4096, "v2 grid4096 NO-epilogue"},
-              {5, 4096, "v5 passthrough 16B->8B"},
-              {6, 4096, "v6 4-deep ILP no-epi"},
-              {7, 4096, "v7 gelu bf16-out no-epi"},
-              {4, 4096, "v4 16B stores+atomics"}};
+    V vs[] = {{0, 4096, "v0 shipped structure"},
+              {2, 4096, "v2 no-atomics (amax live)"},
+              {6, 4096, "v6 ILP4 no-atomics"},
+              {8, 4096, "v8 ILP4+cond-atomic+heur"},
+              {5, 4096, "v5 passthrough"}};
     const int NV = sizeof(vs) / sizeof(vs[0]);
     double best[8];
     for (auto& t : best) t = 1e30;
