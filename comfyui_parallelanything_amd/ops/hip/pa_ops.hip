@@ -340,21 +340,14 @@ __global__ void layer_norm_mod_fp8_kernel(
         float m = 0.f;
         for (int i = 0; i < (int)(blockDim.x / 64); ++i)
             m = fmaxf(m, scratch[i]);
-        atomicMax(reinterpret_cast<unsigned int*>(amax_buf),
-                  __float_as_uint(m));
-        __threadfence();
-        const unsigned int done = atomicAdd(
-            reinterpret_cast<unsigned int*>(&amax_buf[1]), 1u);
-        if (done == gridDim.x - 1) {
-            __threadfence();
-            reinterpret_cast<unsigned int*>(amax_buf)[1] = 0u;
-            const float next = amax_buf[0] * 0.999f;
-            amax_buf[0] = next;
-            if (scale_used != qscale) scale_used[0] = s_entry;
-            qscale[0] = fmaxf(next / 448.f, 1e-12f);
-            __threadfence();
-        }
+        // conditional atomicMax only; decay/scale/snapshot run in the
+        // stream-ordered fp8_scale_finalize_kernel (counter chain removed
+        // — see quant_fp8_bf16_kernel note)
+        if (m > amax_buf[0])
+            atomicMax(reinterpret_cast<unsigned int*>(amax_buf),
+                      __float_as_uint(m));
     }
+    (void)s_entry;
 }
 
 // bf16 fast path: 8-wide gated residual
@@ -439,21 +432,14 @@ __global__ void gelu_fp8_kernel(const bf16* __restrict__ x,
         float m = 0.f;
         for (int i = 0; i < (int)(blockDim.x / 64); ++i)
             m = fmaxf(m, scratch[i]);
-        atomicMax(reinterpret_cast<unsigned int*>(amax_buf),
-                  __float_as_uint(m));
-        __threadfence();
-        const unsigned int done = atomicAdd(
-            reinterpret_cast<unsigned int*>(&amax_buf[1]), 1u);
-        if (done == gridDim.x - 1) {
-            __threadfence();
-            reinterpret_cast<unsigned int*>(amax_buf)[1] = 0u;
-            const float next = amax_buf[0] * 0.999f;
-            amax_buf[0] = next;
-            if (scale_used != scale) scale_used[0] = s_entry;
-            scale[0] = fmaxf(next / 448.f, 1e-12f);
-            __threadfence();
-        }
+        // conditional atomicMax only; decay/scale/snapshot run in the
+        // stream-ordered fp8_scale_finalize_kernel (counter chain removed
+        // — see quant_fp8_bf16_kernel note)
+        if (m > amax_buf[0])
+            atomicMax(reinterpret_cast<unsigned int*>(amax_buf),
+                      __float_as_uint(m));
     }
+    (void)s_entry;
 }
 
 
@@ -741,7 +727,13 @@ __global__ void quant_fp8_bf16_kernel(const bf16* __restrict__ x,
         *reinterpret_cast<unsigned long long*>(&out[i * 8]) =
             *reinterpret_cast<unsigned long long*>(pack);
     }
-    // block amax -> global atomic max (positive fp32 compare as uint)
+    // block amax -> global, CONDITIONAL atomicMax: plain read first, the
+    // atomic only when this block holds a new maximum. The round-2 fp8k
+    // A/B measured the old per-block counter chain (4096 serialized
+    // atomicAdds for last-block detection) at up to ~30% of the whole
+    // kernel; decay/scale/snapshot moved to the stream-ordered
+    // fp8_scale_finalize_kernel launched right after (deterministic:
+    // atomicMax is order-independent, finalize sees the final amax).
     __shared__ float scratch[8];
 #pragma unroll
     for (int off = 32; off > 0; off >>= 1)
@@ -754,34 +746,26 @@ __global__ void quant_fp8_bf16_kernel(const bf16* __restrict__ x,
         float m = 0.f;
         for (int i = 0; i < (int)(blockDim.x / 64); ++i)
             m = fmaxf(m, scratch[i]);
-        atomicMax(reinterpret_cast<unsigned int*>(amax_buf),
-                  __float_as_uint(m));
-        // fused delayed-scaling epilogue (amax_buf[1] = block counter):
-        // the LAST block to finish decays the running amax and writes the
-        // NEXT call's scale — replaces two host-launched micro-kernels per
-        // Linear call. Every other block loaded scale[0] at entry (before
-        // its own counter increment), so the update cannot race a reader.
-        if (fuse_scale) {
-            __threadfence();
-            const unsigned int done = atomicAdd(
-                reinterpret_cast<unsigned int*>(&amax_buf[1]), 1u);
-            if (done == gridDim.x - 1) {
-                __threadfence();
-                reinterpret_cast<unsigned int*>(amax_buf)[1] = 0u;
-                const float next = amax_buf[0] * 0.999f;  // slow decay
-                amax_buf[0] = next;
-                // Snapshot the scale actually USED for this quantization
-                // BEFORE overwriting scale[0] with the next call's value —
-                // _scaled_mm must dequantize with scale_used, not next
-                // (the two differ whenever the activation amax moves).
-                // scale_used may alias scale (legacy callers): skip the
-                // snapshot then, preserving the old in-place behavior.
-                if (scale_used != scale) scale_used[0] = s_entry;
-                scale[0] = fmaxf(next / 448.f, 1e-12f);
-                __threadfence();
-            }
-        }
+        if (m > amax_buf[0])
+            atomicMax(reinterpret_cast<unsigned int*>(amax_buf),
+                      __float_as_uint(m));
     }
+    (void)s_entry;
+    (void)fuse_scale;
+    (void)scale_used;
+}
+
+
+// Stream-ordered delayed-scaling finalize (one thread): snapshot the scale
+// the preceding quantization used, decay the running amax, write the next
+// call's scale. Fixed pointers -> hipGraph-capture safe. amax_buf[1] (the
+// retired block counter) stays zero, preserving the public state contract.
+__global__ void fp8_scale_finalize_kernel(float* scale, float* amax_buf,
+                                          float* scale_used) {
+    if (scale_used != scale) scale_used[0] = scale[0];
+    const float next = amax_buf[0] * 0.999f;
+    amax_buf[0] = next;
+    scale[0] = fmaxf(next / 448.f, 1e-12f);
 }
 
 // ---------------------------------------------------------------------------
@@ -1162,6 +1146,10 @@ at::Tensor layer_norm_mod_fp8(at::Tensor x, at::Tensor scale, at::Tensor shift,
                        (const bf16*)sc.data_ptr(), (const bf16*)sh.data_ptr(),
                        (unsigned char*)out.data_ptr(), n_rows, S, D,
                        (float)eps, qscale.data_ptr<float>(),
+                       amax_buf.data_ptr<float>(),
+                       scale_used.data_ptr<float>());
+    hipLaunchKernelGGL(fp8_scale_finalize_kernel, dim3(1), dim3(1), 0,
+                       cur_stream(), qscale.data_ptr<float>(),
                        amax_buf.data_ptr<float>(),
                        scale_used.data_ptr<float>());
     return out;
@@ -1776,6 +1764,10 @@ at::Tensor gelu_fp8(at::Tensor x, at::Tensor scale, at::Tensor amax_buf,
                        (unsigned char*)out.data_ptr(), total8,
                        scale.data_ptr<float>(), amax_buf.data_ptr<float>(),
                        scale_used.data_ptr<float>());
+    hipLaunchKernelGGL(fp8_scale_finalize_kernel, dim3(1), dim3(1), 0,
+                       cur_stream(), scale.data_ptr<float>(),
+                       amax_buf.data_ptr<float>(),
+                       scale_used.data_ptr<float>());
     return out;
 }
 
@@ -1883,6 +1875,11 @@ at::Tensor quant_fp8(at::Tensor x, at::Tensor scale, at::Tensor amax_buf,
                        (unsigned char*)out.data_ptr(),
                        scale.data_ptr<float>(), amax_buf.data_ptr<float>(),
                        scale_used.data_ptr<float>(), total8, fuse_scale);
+    if (fuse_scale)
+        hipLaunchKernelGGL(fp8_scale_finalize_kernel, dim3(1), dim3(1), 0,
+                           cur_stream(), scale.data_ptr<float>(),
+                           amax_buf.data_ptr<float>(),
+                           scale_used.data_ptr<float>());
     return out;
 }
 
