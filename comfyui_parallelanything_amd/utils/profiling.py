@@ -85,13 +85,22 @@ class StepTimer:
         walls = [s.wall_s for s in self.steps]
         total_imgs = sum(s.images for s in self.steps)
         total_wall = sum(walls)
-        return {
+        out = {
             "steps": len(self.steps),
             "sec_per_it": total_wall / len(self.steps),
             "images_per_s": total_imgs / total_wall if total_wall else 0.0,
             "min_step_s": min(walls),
             "max_step_s": max(walls),
         }
+        if len(self.steps) > 1:
+            # steady state excludes step 0 (cold caches: MIOpen algo
+            # search, RoPE tables, allocator) — the headline-comparable
+            # number for a denoise loop
+            sw = walls[1:]
+            si = sum(s.images for s in self.steps[1:])
+            out["steady_sec_per_it"] = sum(sw) / len(sw)
+            out["steady_images_per_s"] = si / sum(sw) if sum(sw) else 0.0
+        return out
 
     def dump(self, path: str) -> None:
         with open(path, "w") as f:
