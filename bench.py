@@ -225,6 +225,7 @@ def main(argv=None):
                     "sd15": "SD1.5-class UNet",
                     "sd3": "SD3.5-Large-class MMDiT 8B",
                     "wan": "WAN2.2-class video DiT 14B",
+                    "wan5b": "WAN2.2-class dense video DiT 5B",
                     "wan_i2v": "WAN2.2-class I2V video DiT 14B",
                 }[args.model] + (" [TINY DEBUG CONFIG]" if tiny else ""),
                 "global_batch": args.batch,

@@ -144,6 +144,28 @@ def wan_inputs(batch: int, frames: int = 21, h: int = 90, w: int = 160,
     return x, t, ctx, {}
 
 
+def make_wan5b(dev="cpu", dtype=torch.bfloat16, tiny=False):
+    cfg = WanConfig.tiny() if tiny else WanConfig.wan22_5b()
+    torch.manual_seed(0)
+    with torch.device(dev):
+        m = WanDiT(cfg)
+    return m.to(dtype=dtype).eval()
+
+
+def wan5b_inputs(batch: int, frames: int = 21, h: int = 90, w: int = 160,
+                 dev="cpu", dtype=torch.bfloat16, tiny=False,
+                 seed: int = 1234):
+    cfg = WanConfig.tiny() if tiny else WanConfig.wan22_5b()
+    g = torch.Generator(device="cpu").manual_seed(seed)
+    if tiny:
+        frames, h, w = 4, 8, 8
+    x = torch.randn(batch, cfg.in_channels, frames, h, w, generator=g).to(dev, dtype)
+    t = torch.rand(batch, generator=g).to(dev, torch.float32)
+    txt_len = 512 if not tiny else 8
+    ctx = torch.randn(batch, txt_len, cfg.ctx_dim, generator=g).to(dev, dtype)
+    return x, t, ctx, {}
+
+
 def make_wan_i2v(dev="cpu", dtype=torch.bfloat16, tiny=False):
     cfg = WanConfig.tiny_i2v() if tiny else WanConfig.wan22_a14b_i2v()
     torch.manual_seed(0)
@@ -182,5 +204,6 @@ MODELS: Dict[str, Tuple[Callable, Callable]] = {
     "sdxl": (make_sdxl, sdxl_inputs),
     "sd3": (make_sd3, sd3_inputs),
     "wan": (make_wan, wan_inputs),
+    "wan5b": (make_wan5b, wan5b_inputs),
     "wan_i2v": (make_wan_i2v, wan_i2v_inputs),
 }

@@ -105,3 +105,24 @@ def test_bench_zero_size_ranks_world4(tmp_path):
     ref = _single_proc_reference(extra)
     res = _run_world(4, tmp_path, extra)
     assert res["x_checksum"] == pytest.approx(ref["x_checksum"], rel=1e-5)
+
+
+def test_bench_wan_i2v_world2(tmp_path):
+    """wan_i2v through the process-group loop at world 2: 5-D latents and
+    the batch-shaped image_cond kwarg ride scatterv/gatherv."""
+    extra = ["--batch", "4"]
+    base = [a for a in BASE_ARGS]
+    base[base.index("sd15")] = "wan_i2v"
+    # remove --px (wan path takes no px inputs)
+    i = base.index("--px")
+    del base[i:i + 2]
+    import tests.test_bench_loop as me
+
+    old = me.BASE_ARGS
+    me.BASE_ARGS = base
+    try:
+        ref = _single_proc_reference(extra)
+        res = _run_world(2, tmp_path, extra)
+        assert res["x_checksum"] == pytest.approx(ref["x_checksum"], rel=1e-5)
+    finally:
+        me.BASE_ARGS = old
