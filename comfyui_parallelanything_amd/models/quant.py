@@ -150,6 +150,12 @@ class FP8Linear(nn.Module):
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         from .. import ops
 
+        if not x.is_cuda:
+            raise RuntimeError(
+                "FP8Linear is GPU-only (gfx950 _scaled_mm): fp8-quantized "
+                "models cannot run on cpu chain devices — drop the cpu "
+                "entries or skip quantize_fp8"
+            )
         shape = x.shape
         x2 = x.reshape(-1, shape[-1])
         if (x.is_cuda and x.dtype == torch.bfloat16

@@ -348,3 +348,55 @@ def test_fp8_whole_model_error_vs_bf16():
         torch.stack([out.flatten(), ref.flatten()])
     )[0, 1].item()
     assert corr > 0.99, f"fp8 correlation {corr:.4f}"
+
+
+def test_fp8_quant_kernels_deterministic():
+    """Run-to-run bitwise determinism of the delayed-scaling kernels:
+    the conditional atomicMax keeps amax order-independent and the
+    finalize kernel is stream-ordered — identical inputs and state must
+    give identical outputs, scales, and amax."""
+    from comfyui_parallelanything_amd import ops
+
+    if not ops.hip_available("quant_fp8"):
+        pytest.skip("no quant_fp8 in extension")
+    torch.manual_seed(11)
+    x = torch.randn(256, 3072, device="cuda", dtype=torch.bfloat16)
+    sc = torch.randn(1, 3072, device="cuda", dtype=torch.bfloat16) * 0.1
+    sh = torch.randn(1, 3072, device="cuda", dtype=torch.bfloat16) * 0.1
+
+    def run_all():
+        s1 = torch.tensor([0.01], device="cuda")
+        a1 = torch.zeros(2, device="cuda")
+        u1 = torch.zeros(1, device="cuda")
+        q8 = ops.quant_fp8(x, s1, a1, scale_used=u1)
+        s2 = torch.tensor([0.01], device="cuda")
+        a2 = torch.zeros(2, device="cuda")
+        u2 = torch.zeros(1, device="cuda")
+        g8 = ops.gelu_fp8(x, s2, a2, u2)
+        s3 = torch.tensor([0.01], device="cuda")
+        a3 = torch.zeros(2, device="cuda")
+        u3 = torch.zeros(1, device="cuda")
+        l8 = ops.layer_norm_mod_fp8(x.unsqueeze(0), sc, sh, s3, a3, u3)
+        return [t.view(torch.uint8).cpu() for t in (q8, g8, l8)] + [
+            torch.cat([s1, a1, u1, s2, a2, u2, s3, a3, u3]).cpu()
+        ]
+
+    a = run_all()
+    b = run_all()
+    for ta, tb in zip(a, b):
+        assert torch.equal(ta, tb), "fp8 kernel nondeterminism"
+
+
+def test_fp8_linear_cpu_input_raises():
+    """fp8 mode is GPU-only: a cpu input (e.g. a cpu chain device after
+    quantize_fp8) must fail loudly, not corrupt through _scaled_mm."""
+    from comfyui_parallelanything_amd.models.quant import (
+        FP8Linear, _supports_scaled_mm,
+    )
+
+    if not _supports_scaled_mm():
+        pytest.skip("no fp8 _scaled_mm")
+    lin = torch.nn.Linear(64, 64).cuda().to(torch.bfloat16)
+    q = FP8Linear.from_linear(lin)
+    with pytest.raises(RuntimeError, match="GPU-only"):
+        q(torch.randn(4, 64, dtype=torch.bfloat16))
