@@ -163,3 +163,56 @@ def test_cli_tiny_runs(capsys):
           "--steps", "2", "--tiny", "--no-balance"])
     out = capsys.readouterr().out
     assert "images_per_s" in out
+
+
+class ForeignDiT(nn.Module):
+    """A hostile third-party diffusion model, reference-style: lazily
+    caches device-bound tensors under the names clear_flux_caches scrubs
+    (any_device_parallel.py:167-174) and carries a non-picklable attr —
+    the model class the reference's clone ladder exists for."""
+
+    def __init__(self):
+        super().__init__()
+        import threading
+
+        self.inp = nn.Linear(4, 16)
+        self.blk = nn.Linear(16, 16)
+        self.out = nn.Linear(16, 4)
+        self._lock = threading.Lock()   # kills deepcopy
+        self.freqs_cis = None           # device-bound lazy cache
+        self.img_ids = None
+
+    def forward(self, x, timesteps, context=None, **kwargs):
+        B, C, H, W = x.shape
+        if self.freqs_cis is None or self.freqs_cis.device != x.device:
+            self.freqs_cis = torch.linspace(0, 1, 16, device=x.device)
+            self.img_ids = torch.zeros(H * W, device=x.device)
+        h = x.permute(0, 2, 3, 1).reshape(B, H * W, C)
+        h = self.inp(h) + self.freqs_cis + timesteps.reshape(B, 1, 1)
+        h = torch.tanh(self.blk(h)) + 0 * self.img_ids.sum()
+        return self.out(h).reshape(B, H, W, C).permute(0, 3, 1, 2)
+
+
+def test_setup_parallel_foreign_hostile_model():
+    """The node path over a FOREIGN model: structural-clone fallback +
+    cache scrub + DP split must reproduce the single-model output —
+    the reference's core drop-in promise (setup_parallel over arbitrary
+    diffusion_model classes)."""
+    dm = ForeignDiT()
+    x = torch.randn(4, 4, 8, 8)
+    t = torch.rand(4)
+    ref = dm(x, t).clone()              # populates the source's caches
+    assert dm.freqs_cis is not None
+
+    wrapper = FakeModelWrapper(dm)
+    node = ParallelAnything()
+    (out,) = node.setup_parallel(
+        wrapper, _chain(50, 50), auto_vram_balance=False
+    )
+    assert out is wrapper and dm._true_parallel_active
+    y = dm(x, t)                        # DP split through the engine
+    torch.testing.assert_close(y, ref, rtol=1e-5, atol=1e-6)
+    cleanup_parallel_model(dm)
+    assert not getattr(dm, "_true_parallel_active", False)
+    # uninstalled model still runs standalone
+    torch.testing.assert_close(dm(x, t), ref, rtol=1e-5, atol=1e-6)
