@@ -216,3 +216,43 @@ def test_setup_parallel_foreign_hostile_model():
     assert not getattr(dm, "_true_parallel_active", False)
     # uninstalled model still runs standalone
     torch.testing.assert_close(dm(x, t), ref, rtol=1e-5, atol=1e-6)
+
+
+class ForeignBlockDiT(nn.Module):
+    """Foreign model exposing a reference-recognized block list
+    (transformer_blocks, any_device_parallel.py:1156) so batch==1 routes
+    through the block-sharded pipeline."""
+
+    def __init__(self):
+        super().__init__()
+        self.inp = nn.Linear(4, 16)
+        self.transformer_blocks = nn.ModuleList(
+            nn.Linear(16, 16) for _ in range(4)
+        )
+        self.out = nn.Linear(16, 4)
+
+    def forward(self, x, timesteps, context=None, **kwargs):
+        B, C, H, W = x.shape
+        h = self.inp(x.permute(0, 2, 3, 1).reshape(B, H * W, C))
+        h = h + timesteps.reshape(B, 1, 1)
+        for blk in self.transformer_blocks:
+            h = torch.tanh(blk(h))
+        return self.out(h).reshape(B, H, W, C).permute(0, 3, 1, 2)
+
+
+def test_setup_parallel_foreign_pipeline_batch1():
+    """batch==1 on a foreign model with transformer_blocks: the node
+    wires block-sharded pipeline mode and matches the plain forward."""
+    dm = ForeignBlockDiT()
+    x = torch.randn(1, 4, 8, 8)
+    t = torch.rand(1)
+    ref = dm(x, t).clone()
+    wrapper = FakeModelWrapper(dm)
+    node = ParallelAnything()
+    node.setup_parallel(wrapper, _chain(60, 40), auto_vram_balance=False)
+    eng = dm._parallel_engine
+    assert eng.pipeline is not None, "pipeline mode not configured"
+    y = dm(x, t)
+    torch.testing.assert_close(y, ref, rtol=1e-5, atol=1e-6)
+    cleanup_parallel_model(dm)
+    torch.testing.assert_close(dm(x, t), ref, rtol=1e-5, atol=1e-6)
