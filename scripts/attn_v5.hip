@@ -686,6 +686,248 @@ __global__ __launch_bounds__(512, 1) void attn_ws_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// var9 — HALF-STAGGERED schedule: the two block halves run the SAME
+// per-tile phase pair {P1 = QK^T+softmax, P2 = PV+stage} but offset by
+// one phase, so on every SIMD one wave's softmax VALU issues beside its
+// partner's PV MFMAs and vice versa (var8 showed an idle partner
+// uncovers softmax; var5 showed both phases in ONE wave serialize).
+// 2 barriers/tile, 4-slot ring (151.6 KB, 1 block/CU).
+//   seg X: A: P1(t)              B: P2(t) + stage K(t+2)
+//   seg Y: A: P2(t) + stage V(t+2)   B: P1(t+1)
+// ---------------------------------------------------------------------------
+template <int D>
+__global__ __launch_bounds__(512, 1) void attn_st_kernel(
+    const bf16* __restrict__ q, const bf16* __restrict__ k,
+    const bf16* __restrict__ v, bf16* __restrict__ out,
+    int S, int Sk, float scale, int H) {
+    constexpr int KVBLK = 64;
+    constexpr int WAVES = 8;
+    constexpr int KPAD = D + 8;
+    constexpr int VROW = 160;
+    constexpr int KK = D / 16;
+    constexpr int NV = D / 32;
+    constexpr int RING = 4;
+    constexpr int HTHREADS = 256;
+    constexpr int KVECS = (KVBLK * D) / (8 * HTHREADS);
+
+    __shared__ bf16 k_lds[RING * KVBLK * KPAD];
+    __shared__ bf16 v_lds[RING * KVBLK * VROW];
+
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wid = tid >> 6;
+    const int half = wid >> 2;
+    const int hwid = wid & 3;
+    const int htid = (hwid << 6) | lane;
+    const int l32 = lane & 31;
+    const int hi = lane >> 5;
+    const int trb = (hi * 8 + ((lane & 15) >> 2)) * VROW +
+                    16 * (((lane >> 4) & 1) ^ hi) + 4 * (lane & 3);
+
+    const int nq = (S + WAVES * 32 - 1) / (WAVES * 32);
+    const long id = blockIdx.x;
+    const long bh = (id & 7) + 8 * ((id >> 3) / nq);
+    const int qtile = (int)((id >> 3) % nq);
+    const long b = bh / H;
+    const int h = (int)(bh % H);
+    const int q0 = qtile * (WAVES * 32) + half * 128 + hwid * 32;
+    const int ss = H * D;
+
+    const bf16* qp = q + (b * (long)S) * ss + (long)h * D;
+    const bf16* kp = k + (b * (long)Sk) * ss + (long)h * D;
+    const bf16* vp = v + (b * (long)Sk) * ss + (long)h * D;
+    bf16* op = out + (b * (long)S) * ss + (long)h * D;
+
+    bf16x8 qfrag[KK];
+    {
+        const int row = q0 + l32;
+        const int rr = row < S ? row : S - 1;
+#pragma unroll
+        for (int kk = 0; kk < KK; ++kk)
+            qfrag[kk] = *reinterpret_cast<const bf16x8*>(
+                qp + (long)rr * ss + kk * 16 + hi * 8);
+    }
+
+    f32x16 o_acc[NV];
+#pragma unroll
+    for (int n = 0; n < NV; ++n)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) o_acc[n][r] = 0.f;
+    float m_run = -1e30f, l_run = 0.f;
+    const float scale2 = scale * PA_LOG2E;
+    bf16x8 pfrag[4];  // carried from P1 to P2
+
+    auto load_k = [&](int t) {
+        const int slot = t & (RING - 1);
+#pragma unroll
+        for (int i = 0; i < KVECS; ++i) {
+            const int idx = htid + i * HTHREADS;
+            const int row = idx / (D / 8);
+            const int col = (idx % (D / 8)) * 8;
+            const int src = t * KVBLK + row;
+            bf16x8 val = (src < Sk)
+                ? *reinterpret_cast<const bf16x8*>(kp + (long)src * ss + col)
+                : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+            *reinterpret_cast<bf16x8*>(
+                &k_lds[slot * (KVBLK * KPAD) + row * KPAD + col]) = val;
+        }
+    };
+    auto load_v = [&](int t) {
+        const int slot = t & (RING - 1);
+#pragma unroll
+        for (int i = 0; i < KVECS; ++i) {
+            const int idx = htid + i * HTHREADS;
+            const int row = idx / (D / 8);
+            const int col = (idx % (D / 8)) * 8;
+            const int src = t * KVBLK + row;
+            bf16x8 val = (src < Sk)
+                ? *reinterpret_cast<const bf16x8*>(vp + (long)src * ss + col)
+                : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+            *reinterpret_cast<bf16x8*>(
+                &v_lds[slot * (KVBLK * VROW) + row * VROW +
+                       (col ^ ((row & 8) << 1))]) = val;
+        }
+    };
+    auto p1 = [&](int t) {  // QK^T + softmax + P fragments
+        const int buf = t & (RING - 1);
+        const int kv0 = t * KVBLK;
+        f32x16 st[2];
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int kt = 0; kt < 2; ++kt) {
+#pragma unroll
+            for (int r = 0; r < 16; ++r) st[kt][r] = 0.f;
+#pragma unroll
+            for (int kk = 0; kk < KK; ++kk) {
+                bf16x8 afrag = *reinterpret_cast<const bf16x8*>(
+                    &k_lds[buf * (KVBLK * KPAD) + (kt * 32 + l32) * KPAD +
+                           kk * 16 + hi * 8]);
+                st[kt] = mfma32x32x16(afrag, qfrag[kk], st[kt]);
+            }
+        }
+        __builtin_amdgcn_s_setprio(0);
+        float mx = -3e30f;
+#pragma unroll
+        for (int kt = 0; kt < 2; ++kt)
+#pragma unroll
+            for (int r = 0; r < 16; ++r) {
+                const int key = kv0 + kt * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+                const float sv = (key < Sk) ? st[kt][r] : -3e30f;
+                st[kt][r] = sv;
+                mx = fmaxf(mx, sv);
+            }
+        mx = fmaxf(mx, __shfl_xor(mx, 32, 64));
+        const float mnew = fmaxf(m_run, mx * scale2);
+        const float alpha = __builtin_amdgcn_exp2f(m_run - mnew);
+        m_run = mnew;
+        float ps = 0.f;
+#pragma unroll
+        for (int kt = 0; kt < 2; ++kt)
+#pragma unroll
+            for (int r = 0; r < 16; ++r) {
+                const float pv_ =
+                    __builtin_amdgcn_exp2f(fmaf(st[kt][r], scale2, -mnew));
+                st[kt][r] = pv_;
+                ps += pv_;
+            }
+        ps += __shfl_xor(ps, 32, 64);
+        l_run = l_run * alpha + ps;
+        if (alpha != 1.f) {
+#pragma unroll
+            for (int n = 0; n < NV; ++n)
+#pragma unroll
+                for (int r = 0; r < 16; ++r) o_acc[n][r] *= alpha;
+        }
+#pragma unroll
+        for (int c = 0; c < 4; ++c) {
+            const f32x16& sv = st[c >> 1];
+            const int rb = 8 * (c & 1);
+            unsigned int w0 = cvt_pk_bf16(sv[rb + 0], sv[rb + 1]);
+            unsigned int w1 = cvt_pk_bf16(sv[rb + 2], sv[rb + 3]);
+            unsigned int w2 = cvt_pk_bf16(sv[rb + 4], sv[rb + 5]);
+            unsigned int w3 = cvt_pk_bf16(sv[rb + 6], sv[rb + 7]);
+            auto r02 = __builtin_amdgcn_permlane32_swap(w0, w2, false, false);
+            auto r13 = __builtin_amdgcn_permlane32_swap(w1, w3, false, false);
+            unsigned int dd[4] = {(unsigned int)r02[0], (unsigned int)r13[0],
+                                  (unsigned int)r02[1], (unsigned int)r13[1]};
+            pfrag[c] = *reinterpret_cast<bf16x8*>(dd);
+        }
+    };
+    auto p2 = [&](int t) {  // PV from the carried pfrag
+        const int buf = t & (RING - 1);
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int c = 0; c < 4; ++c) {
+#pragma unroll
+            for (int n = 0; n < NV; ++n) {
+                s16x4 alo = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+                    (LDS_P s16x4*)&v_lds[buf * (KVBLK * VROW) + trb +
+                                         c * (16 * VROW) + n * 32]);
+                s16x4 ahi = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+                    (LDS_P s16x4*)&v_lds[buf * (KVBLK * VROW) + trb +
+                                         c * (16 * VROW) + 4 * VROW + n * 32]);
+                bf16x8 va =
+                    __builtin_shufflevector(alo, ahi, 0, 1, 2, 3, 4, 5, 6, 7);
+                o_acc[n] = mfma32x32x16(va, pfrag[c], o_acc[n]);
+            }
+        }
+        __builtin_amdgcn_s_setprio(0);
+    };
+
+    const int n_tiles = (Sk + KVBLK - 1) / KVBLK;
+    // prologue: A stages tile 0, B stages tile 1; then B runs P1(0) while
+    // A stages... B needs K0 (staged by A) -> barrier first.
+    if (half == 0) {
+        load_k(0);
+        load_v(0);
+    } else if (n_tiles > 1) {
+        load_k(1);
+        load_v(1);
+    }
+    __syncthreads();
+    // offset segment: B leads with P1(0); A stages K(2)
+    if (half == 1) p1(0);
+    else if (n_tiles > 2) load_k(2);
+    __syncthreads();
+    for (int t = 0; t < n_tiles; ++t) {
+        // seg X: A P1(t) | B P2(t) + stage V(t+2)
+        if (half == 0) p1(t);
+        else {
+            p2(t);
+            if (t + 2 < n_tiles) load_v(t + 2);
+        }
+        __syncthreads();
+        // seg Y: A P2(t) + stage K(t+3) | B P1(t+1)
+        if (half == 0) {
+            p2(t);
+            if (t + 3 < n_tiles) load_k(t + 3);
+        } else if (t + 1 < n_tiles) {
+            p1(t + 1);
+        }
+        __syncthreads();
+    }
+
+    const int row = q0 + l32;
+    if (row < S) {
+        const float inv_l = (l_run > 0.f) ? 1.f / l_run : 0.f;
+#pragma unroll
+        for (int n = 0; n < NV; ++n)
+#pragma unroll
+            for (int r2 = 0; r2 < 4; ++r2) {
+                const int dim0 = n * 32 + 8 * r2 + 4 * hi;
+                unsigned short pack[4];
+#pragma unroll
+                for (int j = 0; j < 4; ++j)
+                    pack[j] = __bfloat16_as_ushort(
+                        f2bf(o_acc[n][r2 * 4 + j] * inv_l));
+                *reinterpret_cast<unsigned long long*>(
+                    op + (long)row * ss + dim0) =
+                    *reinterpret_cast<unsigned long long*>(pack);
+            }
+    }
+}
+
+// ---------------------------------------------------------------------------
 // var6 — D=64 two-q-tile wave (NOTES r02 item 7): each wave owns 64 q-rows
 // (two 32-row frames). K A-fragments and V tr-reads are shared between the
 // two frames, halving LDS traffic per output row and amortizing the
@@ -1005,7 +1247,7 @@ static void launch(int var, const bf16* q, const bf16* k, const bf16* v,
         hipLaunchKernelGGL((attn_v5_kernel<D, 7>), grid, blk, 0, 0, q, k, v,
                            o, S, S, scale, H, inv_s, s2);
         break;
-    default:
+    case 8:
         // var8: wave-specialized (D=128 only; D=64 -> v5a control)
         if (D == 128)
             hipLaunchKernelGGL((attn_ws_kernel<128>), grid, blk, 0, 0, q, k,
@@ -1013,10 +1255,19 @@ static void launch(int var, const bf16* q, const bf16* k, const bf16* v,
         else
             hipLaunchKernelGGL((attn_v5_kernel<D, 1>), grid, blk, 0, 0, q, k,
                                v, o, S, S, scale, H, inv_s, s2);
+        break;
+    default:
+        // var9: half-staggered phases (D=128 only; D=64 -> v5a control)
+        if (D == 128)
+            hipLaunchKernelGGL((attn_st_kernel<128>), grid, blk, 0, 0, q, k,
+                               v, o, S, S, scale, H);
+        else
+            hipLaunchKernelGGL((attn_v5_kernel<D, 1>), grid, blk, 0, 0, q, k,
+                               v, o, S, S, scale, H, inv_s, s2);
     }
 }
 
-constexpr int NVAR = 9;
+constexpr int NVAR = 10;
 
 template <int D>
 static int check_correct(int B, int H, int S, bool spike) {
