@@ -186,3 +186,23 @@ def test_fp8_model_under_hip_graphs():
         rel = (o.float() - ref.float()).norm() / ref.float().norm()
         assert rel < 0.05, f"fp8 graph replay drifted: {rel:.4f}"
     eng.release()
+
+
+def test_serve_fp8_mode_gpu():
+    """The FastAPI surface with --fp8: quantized engine behind /generate."""
+    from comfyui_parallelanything_amd.models.quant import _supports_scaled_mm
+
+    if not _supports_scaled_mm():
+        pytest.skip("no fp8 _scaled_mm")
+    from fastapi.testclient import TestClient
+
+    from comfyui_parallelanything_amd.serve import create_app
+
+    app = create_app("flux", devices=["cuda:0"], tiny=True, fp8=True)
+    client = TestClient(app)
+    hz = client.get("/healthz").json()
+    assert hz["fp8"] is True
+    r = client.post("/generate", json={"batch": 2, "steps": 2, "seed": 1})
+    assert r.status_code == 200
+    body = r.json()
+    assert body["finite"] is True and body["images_per_s"] > 0
