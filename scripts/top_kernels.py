@@ -1,5 +1,8 @@
 #!/usr/bin/env python3
-"""Summarize a rocprofv3 kernel_stats.csv (args: glob pattern)."""
+"""Summarize a rocprofv3 kernel_stats.csv (args: glob pattern).
+
+NOTE: newer rocprofv3 builds in this image emit rocpd sqlite DBs instead
+of CSVs — use scripts/rocpd_stats.py for those."""
 import csv, glob, sys
 
 pat = sys.argv[1] if len(sys.argv) > 1 else "gpurun_out/prof5/*kernel_stats.csv"
