@@ -163,3 +163,44 @@ def test_dict_of_tensors_cache_cleared():
     m = DictCache()
     clear_replica_caches(m)
     assert m.cache == {}
+
+
+def test_structural_clone_fuzz_attr_types():
+    """Hypothesis-style fuzz (deterministic grid) of the structural path:
+    modules carrying many plain-attr shapes must clone correctly."""
+    import threading
+
+    payloads = [
+        3, 3.5, "s", b"b", None, True,
+        [1, 2, 3], (4, 5), {"a": 1}, {"t": torch.ones(3)},
+        torch.arange(6).reshape(2, 3), [torch.zeros(2), torch.ones(2)],
+        {"nested": {"deep": [torch.full((2,), 7.0)]}},
+        range(5), frozenset({1, 2}),
+    ]
+    for i, payload in enumerate(payloads):
+        class M(nn.Module):
+            def __init__(self):
+                super().__init__()
+                self.lin = nn.Linear(4, 4)
+                self._lock = threading.Lock()   # force structural path
+                self.payload = payload
+
+            def forward(self, x):
+                return self.lin(x)
+
+        torch.manual_seed(i)
+        m = M()
+        x = torch.randn(2, 4)
+        ref = m(x)
+        rep = replicate_module(m, "cpu", force_copy=True)
+        torch.testing.assert_close(rep(x), ref)
+        # tensors in payloads become copies; scalars/strings survive
+        if isinstance(payload, torch.Tensor):
+            assert torch.equal(rep.payload, payload)
+            assert rep.payload.data_ptr() != payload.data_ptr()
+        elif isinstance(payload, (int, float, str, bytes, bool)) or payload is None:
+            assert rep.payload == payload or rep.payload is payload
+        # replica params independent of the source
+        with torch.no_grad():
+            m.lin.weight.add_(1.0)
+        torch.testing.assert_close(rep(x), ref)
