@@ -177,7 +177,10 @@ def _clone_attr(v, target: torch.device, memo: dict, non_blocking: bool):
     if id(v) in memo:
         return memo[id(v)]
     if isinstance(v, torch.Tensor):
-        out = v.to(target, non_blocking=non_blocking)
+        # copy=True: .to() is a no-op for same-device force-copies (the
+        # LoRA clone-even-lead rule) and the replica must never alias
+        # source storage
+        out = v.to(target, non_blocking=non_blocking, copy=True)
         memo[id(v)] = out
         return out
     if isinstance(v, nn.Module):
