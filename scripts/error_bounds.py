@@ -1,3 +1,5 @@
+import os, sys
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import copy, torch
 from comfyui_parallelanything_amd.models.registry import MODELS
 from comfyui_parallelanything_amd.models.quant import quantize_fp8
