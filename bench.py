@@ -230,7 +230,17 @@ def main(argv=None):
                 }[args.model] + (" [TINY DEBUG CONFIG]" if tiny else ""),
                 "global_batch": args.batch,
                 "resolution": args.px,
-                "seq_len": (args.px // 16) ** 2 + 512 if args.model == "flux" else None,
+                "seq_len": {
+                    # tokens entering self-attention at the named config
+                    "flux": (args.px // 16) ** 2 + 512,
+                    "sd3": (args.px // 16) ** 2 + 154,
+                    "zimage": (args.px // 16) ** 2 + 64,
+                    "sdxl": (args.px // 16) ** 2,   # deepest-level grid
+                    "sd15": (args.px // 16) ** 2,
+                    "wan": 21 * 45 * 80,            # 720p x 21f, patch (1,2,2)
+                    "wan5b": 21 * 45 * 80,
+                    "wan_i2v": 21 * 45 * 80,
+                }.get(args.model),
                 "parallelism": f"dp{n}" + (
                     f" weighted {args.weights}" if args.weights else ""),
             },
