@@ -409,6 +409,8 @@ class ParallelEngine:
         self.replicas.clear()
         self.streams.clear()
         self.pipeline = None
+        self.balancer = None
+        self._pending_times.clear()
         if self.graphs is not None:
             self.graphs.clear()
         if torch.cuda.is_available():
