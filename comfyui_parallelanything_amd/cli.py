@@ -35,7 +35,9 @@ def main(argv=None):
     ap.add_argument("--batch", type=int, default=8)
     ap.add_argument("--px", type=int, default=1024)
     ap.add_argument("--steps", type=int, default=10)
-    ap.add_argument("--dtype", default="bf16", choices=["bf16", "fp16", "fp32"])
+    ap.add_argument("--dtype", default="bf16",
+                    choices=["bf16", "fp16", "fp32", "fp8"],
+                    help="fp8 = e4m3fn serving mode (GPU chains only)")
     ap.add_argument("--tiny", action="store_true")
     ap.add_argument("--no-split", dest="split", action="store_false")
     ap.add_argument("--no-balance", dest="balance", action="store_false")
@@ -59,7 +61,7 @@ def main(argv=None):
     if len(pcts) != len(devices):
         raise SystemExit("--percent count must match --devices count")
     dtype = {"bf16": torch.bfloat16, "fp16": torch.float16,
-             "fp32": torch.float32}[args.dtype]
+             "fp32": torch.float32, "fp8": torch.bfloat16}[args.dtype]
     tiny = args.tiny or not torch.cuda.is_available()
     if tiny and dtype != torch.float32:
         dtype = torch.float32
@@ -80,6 +82,11 @@ def main(argv=None):
         path, _, s = args.lora.partition(":")
         n = merge_lora_file(model, path, scale=float(s) if s else 1.0)
         print(f"[lora] merged {n} modules from {path}")
+
+    if args.dtype == "fp8":
+        from .models.quant import quantize_fp8
+
+        quantize_fp8(model)
 
     engine = ParallelEngine(chain, workload_split=args.split,
                             auto_vram_balance=args.balance,
